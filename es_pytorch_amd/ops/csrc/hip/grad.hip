@@ -1,0 +1,66 @@
+// K5 — gather-GEMV gradient reconstruction (the flagship op).
+//
+//   g[t] = sum_p fits[p] * table[offsets[p] + t]        t in [0, n_params)
+//
+// The reference does this as batched numpy dots over copied noise rows
+// (src/utils/utils.py:29-39, batch_size=500 rows). Here the rows are never
+// copied: each thread owns ELEMS consecutive output elements (strided across
+// the block for coalescing) and walks the whole population, reading each
+// row's slice directly from the HBM table. (fits, offset) pairs are staged
+// through LDS in tiles. Memory-bound: pop x n_params x 4B table reads.
+#include "common.h"
+
+#define ES_GRAD_TILE 256
+#define ES_GRAD_ELEMS 4
+
+__global__ void __launch_bounds__(256)
+grad_gather_kernel(float* __restrict__ g, const float* __restrict__ table,
+                   const float* __restrict__ fits, const int64_t* __restrict__ offsets,
+                   int64_t n_pop, int64_t n_params) {
+  __shared__ float s_fit[ES_GRAD_TILE];
+  __shared__ int64_t s_off[ES_GRAD_TILE];
+
+  const int64_t chunk = (int64_t)blockDim.x * ES_GRAD_ELEMS;
+  const int64_t base = blockIdx.x * chunk;
+  float acc[ES_GRAD_ELEMS];
+#pragma unroll
+  for (int k = 0; k < ES_GRAD_ELEMS; ++k) acc[k] = 0.0f;
+
+  for (int64_t p0 = 0; p0 < n_pop; p0 += ES_GRAD_TILE) {
+    const int tile = (int)min((int64_t)ES_GRAD_TILE, n_pop - p0);
+    __syncthreads();
+    for (int i = threadIdx.x; i < tile; i += blockDim.x) {
+      s_fit[i] = fits[p0 + i];
+      s_off[i] = offsets[p0 + i];
+    }
+    __syncthreads();
+    for (int i = 0; i < tile; ++i) {
+      const float f = s_fit[i];
+      const float* row = table + s_off[i] + base;
+#pragma unroll
+      for (int k = 0; k < ES_GRAD_ELEMS; ++k) {
+        const int64_t t = base + k * blockDim.x + threadIdx.x;
+        if (t < n_params) acc[k] = fmaf(f, row[k * blockDim.x + threadIdx.x], acc[k]);
+      }
+    }
+  }
+
+#pragma unroll
+  for (int k = 0; k < ES_GRAD_ELEMS; ++k) {
+    const int64_t t = base + k * blockDim.x + threadIdx.x;
+    if (t < n_params) g[t] = acc[k];
+  }
+}
+
+extern "C" int es_grad_gather(void* g, const void* table, const void* fits,
+                              const void* offsets, int64_t n_pop, int64_t n_params,
+                              void* stream) {
+  const int threads = 256;
+  const int64_t chunk = (int64_t)threads * ES_GRAD_ELEMS;
+  const int blocks = (int)((n_params + chunk - 1) / chunk);
+  grad_gather_kernel<<<dim3(blocks), dim3(threads), 0, (hipStream_t)stream>>>(
+      (float*)g, (const float*)table, (const float*)fits, (const int64_t*)offsets, n_pop,
+      n_params);
+  ES_CHECK_LAUNCH();
+  return 0;
+}
