@@ -1,0 +1,209 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: Humanoid-v2 MLP(256,256) objective-ES (BASELINE.json).
+
+Measures whole-job env-steps/sec (and generations/sec) of the GPU engine:
+population-batched synthetic Humanoid rollouts, HBM noise table, HIP kernels,
+RCCL all-gather across ranks. Weak scaling: policies_per_gen = 1280 per GPU
+(10240 at 8 GPUs = the BASELINE "pop=10k on 8 GPUs" config).
+
+Launch (driver contract):
+  python bench.py --gpus 1 --steps K --warmup W                 # single GPU
+  python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+      --master-addr 127.0.0.1 bench.py --gpus N ...             # one rank/GPU
+
+A "step" is one ES generation. Data is synthetic (seeded latent-dynamics env
+of Humanoid-v2's 376-obs/17-action shape; no physics engine or datasets exist
+offline) with random-init weights; compute dtype bf16 (weights/forward) with
+fp32 master params and fp64 fitness exchange.
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import time
+
+import numpy as np
+import torch
+
+
+def main():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=5, help="timed generations")
+    p.add_argument("--warmup", type=int, default=2, help="untimed warmup generations")
+    p.add_argument("--pop-per-gpu", type=int, default=1280)
+    p.add_argument("--max-steps", type=int, default=1000, help="env steps per episode")
+    p.add_argument("--env", type=str, default="Humanoid-v2")
+    p.add_argument("--layers", type=int, nargs="*", default=[256, 256])
+    p.add_argument("--tbl-size", type=int, default=250_000_000)
+    p.add_argument("--no-graph", action="store_true")
+    p.add_argument("--cpu", action="store_true", help="debug: run the engine on CPU")
+    args = p.parse_args()
+
+    from es_pytorch_amd.config import AttrDict
+    from es_pytorch_amd.core.engine import GpuEngine
+    from es_pytorch_amd.core.noisetable import NoiseTable
+    from es_pytorch_amd.core.policy import Policy
+    from es_pytorch_amd.envs import make_batched
+    from es_pytorch_amd.nn.nn import FeedForward
+    from es_pytorch_amd.nn.optimizers import Adam
+    from es_pytorch_amd.parallel.comm import init_comm, seed_all
+    from es_pytorch_amd.utils.rankers import CenteredRanker
+
+    use_cuda = torch.cuda.is_available() and not args.cpu
+    comm = init_comm()
+    world = comm.size
+    if world > 1 and args.gpus != world:
+        args.gpus = world
+    device = comm.device if use_cuda else torch.device("cpu")
+
+    rs, my_seed, global_seed = seed_all(comm, [1000 + r for r in range(world)])
+
+    ppg = args.pop_per_gpu * world
+    cfg = AttrDict({
+        "env": {"name": args.env, "max_steps": args.max_steps},
+        "noise": {"tbl_size": args.tbl_size, "std": 0.02},
+        "policy": {"layer_sizes": list(args.layers), "ac_std": 0.01, "l2coeff": 0.005,
+                   "lr": 0.01, "ob_clip": 5, "save_obs_chance": 0.01},
+        "general": {"name": "bench", "policies_per_gen": ppg, "batch_size": 500,
+                    "seed": global_seed},
+    })
+
+    B = 2 * (ppg // world // 2) + 1
+    # fixed-horizon synthetic rollouts: every counted env step is a fully
+    # computed batched forward + dynamics step (stable, honest throughput)
+    env = make_batched(cfg.env.name, B, device, max_steps=args.max_steps,
+                       terminate_on_fall=False)
+    nn = FeedForward(cfg.policy.layer_sizes, torch.nn.Tanh(), env, cfg.policy.ac_std,
+                     cfg.policy.ob_clip)
+    policy = Policy(nn, cfg.noise.std, Adam(len(Policy.get_flat(nn)), cfg.policy.lr))
+
+    if use_cuda and not os.path.exists(os.path.join(os.path.dirname(
+            os.path.abspath(__file__)), "es_pytorch_amd", "ops", "_hip_ops.so")):
+        raise RuntimeError("HIP ops library missing on a GPU machine — build first")
+
+    tbl = args.tbl_size if use_cuda else min(args.tbl_size, 2_000_000)
+    nt = NoiseTable.create_shared(comm, tbl, len(policy), seed=cfg.general.seed,
+                                  device=device)
+
+    if use_cuda:
+        engine = GpuEngine(cfg, comm, policy, nt, env, rs, objective="reward",
+                           use_graph=not args.no_graph)
+    else:
+        engine = _CpuRefEngine(cfg, comm, policy, nt, env, rs)
+    ranker = CenteredRanker()
+
+    for _ in range(args.warmup):
+        engine.step(ranker)
+
+    comm.barrier()
+    if use_cuda:
+        torch.cuda.synchronize(device)
+    t0 = time.perf_counter()
+    steps_done = 0
+    for _ in range(args.steps):
+        engine.step(ranker)
+        steps_done += engine.timings["env_steps"]
+    comm.barrier()
+    if use_cuda:
+        torch.cuda.synchronize(device)
+    elapsed = time.perf_counter() - t0
+
+    # MAX elapsed over ranks = whole-job wall time
+    elapsed = comm.allreduce_scalar(elapsed) if world == 1 else max(
+        comm.allgather_obj(elapsed))
+    # env_steps from engine are already whole-job (allreduced per gen)
+    env_steps_per_sec = steps_done / elapsed
+    gens_per_sec = args.steps / elapsed
+
+    if comm.rank == 0:
+        out = {
+            "metric": "env-steps/sec (whole node), Humanoid-v2 ES",
+            "value": round(env_steps_per_sec, 1),
+            "unit": "env-steps/s",
+            "n_gpus": args.gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(1000.0 * elapsed / args.steps, 2),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16",
+            "data": "synthetic",
+            "config": {
+                "model": f"MLP{tuple(args.layers)}",
+                "env": args.env,
+                "global_batch": ppg,
+                "policies_per_gen": ppg,
+                "episode_len": args.max_steps,
+                "seq_len": args.max_steps,
+                "parallelism": f"dp{args.gpus}",
+                "gens_per_sec": round(gens_per_sec, 3),
+                "noise_table_elems": tbl,
+                "n_params": len(policy),
+            },
+        }
+        print(json.dumps(out))
+    return 0
+
+
+class _CpuRefEngine:
+    """Tiny CPU stand-in so bench.py --cpu runs without a GPU (debug only)."""
+
+    def __init__(self, cfg, comm, policy, nt, env, rs):
+        from es_pytorch_amd.core import engine as _e
+        self.cfg, self.comm, self.policy, self.nt, self.env, self.rs = \
+            cfg, comm, policy, nt, env, rs
+        self.timings = {}
+        self.pairs = cfg.general.policies_per_gen // comm.size // 2
+        self.B = env.batch
+        self.perm = _e.forward_perm(policy._module.layer_dims())
+
+    def step(self, ranker):
+        import torch as T
+        t0 = time.perf_counter()
+        cfg, env, policy = self.cfg, self.env, self.policy
+        offs = self.nt.sample_idxs(self.rs, self.pairs)
+        n = len(policy)
+        theta = T.from_numpy(policy.flat_params)
+        signs = np.concatenate([np.ones(self.pairs), -np.ones(self.pairs), [0.0]])
+        offs_full = np.concatenate([offs, offs, [0]]).astype(np.int64)
+        W = T.stack([theta + float(s) * policy.std * self.nt.noise[o:o + n]
+                     for s, o in zip(signs, offs_full)])
+        obs = env.reset(0)
+        rew_tot = T.zeros(self.B)
+        alive = T.ones(self.B)
+        steps = T.zeros(())
+        dims = policy._module.layer_dims()
+        mean = T.zeros(env.ob_dim)
+        std = T.ones(env.ob_dim)
+        for t in range(cfg.env.max_steps):
+            x = T.clamp((obs - mean) / std, -5, 5)
+            off = 0
+            for li, (I, O) in enumerate(zip(dims[:-1], dims[1:])):
+                w = W[:, off:off + I * O].reshape(self.B, O, I)
+                b = W[:, off + I * O:off + I * O + O]
+                x = T.tanh(T.einsum("boi,bi->bo", w, x) + b)
+                off += I * O + O
+            obs, rew, done = env.step(x)
+            rew_tot += rew * alive
+            steps += alive.sum()
+            alive *= (1 - done.float())
+        rows = np.zeros((self.pairs, 3))
+        rows[:, 0] = rew_tot[:self.pairs].numpy()
+        rows[:, 1] = rew_tot[self.pairs:2 * self.pairs].numpy()
+        rows[:, 2] = offs
+        allr = self.comm.allgather_rows(T.from_numpy(rows)).numpy()
+        ranker.rank(allr[:, :1], allr[:, 1:2], allr[:, 2])
+        from es_pytorch_amd.utils.utils import scale_noise
+        g = scale_noise(ranker.ranked_fits, ranker.noise_inds, self.nt, n,
+                        cfg.general.batch_size).numpy() / ranker.n_fits_ranked
+        policy.optim_step(cfg.policy.l2coeff * policy.flat_params - g)
+        total = float(self.comm.allreduce_scalar(float(steps.item())))
+        self.timings = {"env_steps": total, "gen_s": time.perf_counter() - t0}
+
+
+if __name__ == "__main__":
+    sys.exit(main())

@@ -1,0 +1,368 @@
+"""GpuEngine — whole-generation population-batched ES on one GPU per rank.
+
+This is the MI355X-native replacement for the reference's hot loop
+(``src/core/es.py:54-81``: E sequential pheno+rollout pairs per CPU process).
+One rank = one GPU; per generation:
+
+1. sample the rank's noise offsets (numpy RandomState, reference-compatible
+   index semantics, ``noisetable.py:37-40``) — one batched draw;
+2. HIP pheno kernel materializes ALL local members' perturbed parameters as
+   bf16 blobs in HBM: rows [+noise pairs | -noise pairs | noiseless slot]
+   (the noiseless evaluation of ``es.py:48`` rides in the same batch);
+3. the rollout loop — fused HIP MLP forward over the whole population +
+   batched env dynamics — is captured once into a hipGraph and replayed
+   every generation (launch-overhead-free inner loop);
+   finished episodes stay in the batch under an ``alive`` mask with their
+   rewards/behaviour frozen, matching the reference's break-on-done
+   (``gym_runner.py:63-64``);
+4. (fit+, fit-, idx) fp64 triples go through ONE RCCL all_gather over xGMI
+   (replaces the replicated Alltoall, ``es.py:84-95``); ranking runs
+   redundantly on every rank (reference README.md:10-12 — parameters are
+   never communicated);
+5. the HIP gather-GEMV reconstructs the gradient straight from the HBM noise
+   table and the fused Adam/SGD kernel updates theta in place.
+
+Parameter layout: device theta lives in FORWARD layout (per layer: W^T, b)
+so pheno/forward/grad are all coalesced; a precomputed permutation maps it
+to the reference's flat state_dict layout at the host boundary (checkpoint
+save/load keeps the reference's pickle format, ``policy.py:37-47``).
+"""
+from __future__ import annotations
+
+import time
+from typing import List, Optional, Tuple
+
+import numpy as np
+import torch
+
+from es_pytorch_amd import ops
+from es_pytorch_amd.core.noisetable import NoiseTable
+from es_pytorch_amd.core.policy import Policy
+from es_pytorch_amd.envs.base import BatchedEnv
+from es_pytorch_amd.nn.obstat import ObStat
+from es_pytorch_amd.nn.optimizers import SGD, Adam, Optimizer
+from es_pytorch_amd.parallel.comm import Comm
+from es_pytorch_amd.rollout.results import RewardResult
+from es_pytorch_amd.utils.novelty import novelty_batch
+from es_pytorch_amd.utils.rankers import Ranker
+
+
+def forward_perm(dims: List[int]) -> torch.Tensor:
+    """Map forward-layout element index -> flat state_dict index.
+
+    flat (reference ``policy.py:33-35``): per layer, W (O, I) row-major then
+    b (O). forward: per layer, W^T (I, O) row-major then b (O).
+    """
+    perm = []
+    flat_off = 0
+    for I, O in zip(dims[:-1], dims[1:]):
+        w = np.arange(O * I, dtype=np.int64).reshape(O, I)  # flat W indices
+        perm.append((flat_off + w.T).reshape(-1))           # (I, O) order
+        flat_off += O * I
+        perm.append(np.arange(O, dtype=np.int64) + flat_off)
+        flat_off += O
+    return torch.from_numpy(np.concatenate(perm))
+
+
+class _NoiselessResult(RewardResult):
+    """Adapter presenting the engine's noiseless slot to the Reporter API."""
+
+    def __init__(self, reward: float, position, steps: int, ob_shape):
+        super().__init__([reward], [float(position[0]), float(position[1]),
+                                    float(position[2])] * 2,
+                         np.zeros((1,) + tuple(ob_shape)), steps)
+
+
+class GpuEngine:
+    def __init__(self, cfg, comm: Comm, policy: Policy, nt: NoiseTable, env: BatchedEnv,
+                 rs: np.random.RandomState, objective: str = "reward",
+                 use_graph: bool = True, novelty_k: int = 10):
+        self.cfg = cfg
+        self.comm = comm
+        self.policy = policy
+        self.nt = nt
+        self.env = env
+        self.rs = rs
+        self.objective = objective
+        self.device = nt.noise.device
+        self.use_graph = use_graph and self.device.type == "cuda"
+        self.novelty_k = novelty_k
+        self.archive: Optional[torch.Tensor] = None  # (N, 2) device tensor for ns/nsr
+
+        ppg = cfg.general.policies_per_gen
+        assert ppg % comm.size == 0 and (ppg / comm.size) % 2 == 0, \
+            "policies_per_gen must split into antithetic pairs per rank"
+        self.pairs = int(ppg // comm.size // 2)
+        self.B = 2 * self.pairs + 1  # [+pairs | -pairs | noiseless]
+        assert env.batch == self.B, f"env batch {env.batch} != engine batch {self.B}"
+
+        self.dims = policy._module.layer_dims()
+        self.n = int(np.sum([I * O + O for I, O in zip(self.dims[:-1], self.dims[1:])]))
+        assert self.n == len(policy), (self.n, len(policy))
+        self.perm = forward_perm(self.dims).to(self.device)  # fwd idx -> flat idx
+        self.dims_arr = (np.array(self.dims, dtype=np.int32))
+
+        d = self.device
+        flat = torch.from_numpy(policy.flat_params).to(d)
+        self.theta = flat[self.perm].contiguous()            # forward layout fp32
+        self.m = torch.zeros(self.n, dtype=torch.float32, device=d)
+        self.v = torch.zeros(self.n, dtype=torch.float32, device=d)
+        self._load_optim_state()
+
+        self.weights = torch.empty((self.B, self.n), dtype=torch.bfloat16, device=d)
+        self.offsets = torch.zeros(self.B, dtype=torch.int64, device=d)
+        self.signs = torch.cat([torch.ones(self.pairs), -torch.ones(self.pairs),
+                                torch.zeros(1)]).to(d)
+        self.grad = torch.empty(self.n, dtype=torch.float32, device=d)
+        self.seed_dev = torch.zeros(1, dtype=torch.int64, device=d)
+
+        D = env.ob_dim
+        self.obmean = torch.zeros(D, dtype=torch.float32, device=d)
+        self.obstd = torch.ones(D, dtype=torch.float32, device=d)
+        self._push_obstat()
+
+        # rollout state buffers (persistent; written in-place inside the graph)
+        self.actions = torch.empty((self.B, env.ac_dim), dtype=torch.float32, device=d)
+        self.alive = torch.ones(self.B, dtype=torch.float32, device=d)
+        self.rew_total = torch.zeros(self.B, dtype=torch.float32, device=d)
+        self.steps_total = torch.zeros((), dtype=torch.float32, device=d)
+        self.member_steps = torch.zeros(self.B, dtype=torch.float32, device=d)
+        self.behv = torch.zeros((self.B, 3), dtype=torch.float32, device=d)
+        self.save_mask = torch.zeros(self.B, dtype=torch.float32, device=d)
+        self.ob_sum = torch.zeros(D, dtype=torch.float64, device=d)
+        self.ob_sumsq = torch.zeros(D, dtype=torch.float64, device=d)
+        self.ob_count = torch.zeros((), dtype=torch.float64, device=d)
+        self.obs_buf = torch.zeros((self.B, D), dtype=torch.float32, device=d)
+
+        self.max_steps = int(cfg.env.max_steps)
+        self.gen = 0
+        self._graph: Optional[torch.cuda.CUDAGraph] = None
+        self.timings = {}
+
+    # ------------------------------------------------------------------ ops
+    def _stream(self):
+        return torch.cuda.current_stream(self.device).cuda_stream if \
+            self.device.type == "cuda" else None
+
+    def _pheno(self):
+        std = float(self.policy.std)
+        ops.check(ops.hip().es_pheno_bf16(
+            self.weights.data_ptr(), self.theta.data_ptr(), self.nt.noise.data_ptr(),
+            self.offsets.data_ptr(), self.signs.data_ptr(), self.B, self.n, std,
+            self._stream()), "es_pheno_bf16")
+
+    def _forward(self, obs: torch.Tensor, salt: int):
+        ac_std = float(getattr(self.policy._module, "_action_std", 0.0))
+        ops.check(ops.hip().es_mlp_fwd(
+            self.actions.data_ptr(), obs.data_ptr(), self.weights.data_ptr(),
+            self.obmean.data_ptr(), self.obstd.data_ptr(),
+            self.dims_arr.ctypes.data, len(self.dims_arr), self.seed_dev.data_ptr(),
+            salt, self.B, float(self.policy._module.ob_clip), ac_std, self.n, 1,
+            self._stream()), "es_mlp_fwd")
+        return self.actions
+
+    # ------------------------------------------------------------- rollout
+    def _step_body(self, t: int):
+        a = self._forward(self.obs_buf, salt=t + 1)
+        ob, rew, done = self.env.step(a)
+        alive = self.alive
+        self.rew_total.add_(rew * alive)
+        self.steps_total.add_(alive.sum())
+        self.member_steps.add_(alive)
+        am = alive.bool().unsqueeze(1)
+        self.behv.copy_(torch.where(am, self.env.positions, self.behv))
+        # per-gen observation statistics, save_mask-gated like the reference's
+        # save_obs_chance episodes (simple_example.py:38, obstat fed at es.py:73-74)
+        w = alive * self.save_mask
+        self.ob_sum.add_((ob * w.unsqueeze(1)).sum(0).double())
+        self.ob_sumsq.add_((ob * ob * w.unsqueeze(1)).sum(0).double())
+        self.ob_count.add_(w.sum().double())
+        self.alive.mul_(1.0 - done.float())
+        self.obs_buf.copy_(ob)
+
+    def _rollout(self):
+        if self.use_graph:
+            if self._graph is None:
+                # warmup (lazy inits must happen outside capture), then capture once
+                s = torch.cuda.Stream(self.device)
+                s.wait_stream(torch.cuda.current_stream(self.device))
+                with torch.cuda.stream(s):
+                    for t in range(3):
+                        self._step_body(t)
+                torch.cuda.current_stream(self.device).wait_stream(s)
+                # warmup dirtied the rollout state: restore it before capture
+                self._reset_rollout_state()
+                self.obs_buf.copy_(self.env.reset(self._gen_seed()))
+                self._graph = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(self._graph):
+                    for t in range(self.max_steps):
+                        self._step_body(t)
+            self._graph.replay()
+            return
+        for t in range(self.max_steps):
+            self._step_body(t)
+
+    def _reset_rollout_state(self):
+        self.alive.fill_(1.0)
+        self.rew_total.zero_()
+        self.steps_total.zero_()
+        self.member_steps.zero_()
+        self.behv.zero_()
+        self.ob_sum.zero_()
+        self.ob_sumsq.zero_()
+        self.ob_count.zero_()
+
+    def _gen_seed(self) -> int:
+        # distinct env variations per generation, identical across ranks only
+        # for the matrices (seeded in env ctor); initial states vary per rank
+        return (self.gen * 1000003 + self.comm.rank * 7919) & 0x7FFFFFFF
+
+    # ------------------------------------------------------------- fitness
+    def _fitnesses(self) -> torch.Tensor:
+        """(B, O) fitness per member from accumulated rollout state."""
+        if self.objective == "reward":
+            return self.rew_total.unsqueeze(1)
+        if self.objective == "mean_reward":
+            return (self.rew_total / torch.clamp(self.member_steps, min=1.0)).unsqueeze(1)
+        if self.objective == "dist":
+            return self.behv[:, :2].norm(dim=1, keepdim=True)
+        if self.objective == "xdist":
+            return self.behv[:, :1]
+        if self.objective in ("ns", "nsr"):
+            assert self.archive is not None, "set engine.archive before ns/nsr generations"
+            nov = novelty_batch(self.behv[:, :2], self.archive, self.novelty_k)
+            if self.objective == "ns":
+                return nov.unsqueeze(1)
+            return torch.stack([self.rew_total, nov], dim=1)
+        raise ValueError(f"unknown objective {self.objective!r}")
+
+    # ---------------------------------------------------------------- step
+    def step(self, ranker: Ranker, reporter=None) -> Tuple[_NoiselessResult, ObStat]:
+        """Run one full generation; mirrors ``es.step`` semantics (``es.py:23-51``)."""
+        t0 = time.perf_counter()
+        cfg = self.cfg
+        offs = self.nt.sample_idxs(self.rs, self.pairs)
+        self.offsets[:self.pairs].copy_(torch.from_numpy(offs).to(self.device))
+        self.offsets[self.pairs:2 * self.pairs].copy_(self.offsets[:self.pairs])
+        self.seed_dev.fill_(int(self.rs.randint(0, 2 ** 31)))
+        chance = float(cfg.policy.get("save_obs_chance", 1.0))
+        sm = (self.rs.random_sample(self.B) < chance).astype(np.float32)
+        sm[-1] = 0.0  # noiseless slot never feeds the gen obstat (reference es.py:48 is separate)
+        self.save_mask.copy_(torch.from_numpy(sm).to(self.device))
+
+        self._pheno()
+        self._reset_rollout_state()
+        self.obs_buf.copy_(self.env.reset(self._gen_seed()))
+        t1 = time.perf_counter()
+        self._rollout()
+        fits = self._fitnesses()
+
+        # -- share (fit+, fit-, idx) triples: ONE RCCL all_gather over xGMI
+        O = fits.shape[1]
+        rows = torch.empty((self.pairs, 2 * O + 1), dtype=torch.float64, device=self.device)
+        rows[:, :O] = fits[:self.pairs].double()
+        rows[:, O:2 * O] = fits[self.pairs:2 * self.pairs].double()
+        rows[:, -1] = self.offsets[:self.pairs].double()
+        if self.device.type == "cuda":
+            torch.cuda.synchronize(self.device)
+        t2 = time.perf_counter()
+        all_rows = self.comm.allgather_rows(rows).cpu().numpy()
+        steps = int(self.comm.allreduce_scalar(float(self.steps_total.item())))
+        t3 = time.perf_counter()
+
+        # -- identical redundant ranking on every rank (host; pop-sized)
+        pos, neg = all_rows[:, :O], all_rows[:, O:2 * O]
+        inds = all_rows[:, -1]
+        ranker.rank(pos, neg, inds)
+
+        # -- gradient + update on device (HIP gather-GEMV + fused Adam/SGD)
+        rf = torch.from_numpy(np.ascontiguousarray(ranker.ranked_fits,
+                                                   dtype=np.float32)).to(self.device)
+        ri = torch.from_numpy(np.ascontiguousarray(ranker.noise_inds,
+                                                   dtype=np.int64)).to(self.device)
+        ops.check(ops.hip().es_grad_gather(
+            self.grad.data_ptr(), self.nt.noise.data_ptr(), rf.data_ptr(), ri.data_ptr(),
+            rf.numel(), self.n, self._stream()), "es_grad_gather")
+        self._optim_step(float(ranker.n_fits_ranked), float(cfg.policy.l2coeff))
+        self.sync_host(light=True)
+        t4 = time.perf_counter()
+
+        # -- per-gen obstat -> merged across ranks (packed all_reduce)
+        gen_obstat = ObStat(self.env.observation_space.shape, 0)
+        gen_obstat.inc(self.ob_sum.cpu().numpy(), self.ob_sumsq.cpu().numpy(),
+                       float(self.ob_count.item()))
+        gen_obstat.dist_inc(self.comm)
+
+        nl_rew = float(self.rew_total[-1].item())
+        nl_pos = self.behv[-1].cpu().numpy()
+        noiseless = _NoiselessResult(nl_rew, nl_pos, steps, self.env.observation_space.shape)
+        if reporter is not None:
+            reporter.log_gen(np.concatenate([pos, neg]), noiseless, self.policy, steps)
+
+        self.gen += 1
+        self.timings = {"pheno_s": t1 - t0, "rollout_s": t2 - t1, "comm_s": t3 - t2,
+                        "update_s": t4 - t3, "gen_s": time.perf_counter() - t0,
+                        "env_steps": steps}
+        return noiseless, gen_obstat
+
+    # ------------------------------------------------------------- update
+    def _optim_step(self, n_ranked: float, l2: float):
+        opt = self.policy.optim
+        opt.t += 1
+        gscale = 1.0 / n_ranked
+        if isinstance(opt, Adam):
+            a = opt.lr * np.sqrt(1 - opt.beta2 ** opt.t) / (1 - opt.beta1 ** opt.t)
+            ops.check(ops.hip().es_adam_step(
+                self.theta.data_ptr(), self.m.data_ptr(), self.v.data_ptr(),
+                self.grad.data_ptr(), self.n, float(a), float(opt.beta1), float(opt.beta2),
+                float(opt.epsilon), l2, gscale, self._stream()), "es_adam_step")
+        elif isinstance(opt, SGD):
+            ops.check(ops.hip().es_sgd_step(
+                self.theta.data_ptr(), self.v.data_ptr(), self.grad.data_ptr(), self.n,
+                float(opt.lr), float(opt.momentum), l2, gscale, self._stream()),
+                "es_sgd_step")
+        else:
+            raise TypeError(f"GpuEngine supports Adam/SGD, got {type(opt).__name__}")
+
+    def _load_optim_state(self):
+        opt: Optimizer = self.policy.optim
+        if isinstance(opt, Adam) and np.any(opt.m):
+            self.m.copy_(torch.from_numpy(opt.m).to(self.device)[self.perm])
+            self.v.copy_(torch.from_numpy(opt.v).to(self.device)[self.perm])
+        elif isinstance(opt, SGD) and np.any(opt.v):
+            self.v.copy_(torch.from_numpy(opt.v).to(self.device)[self.perm])
+
+    def _push_obstat(self):
+        self.obmean.copy_(torch.from_numpy(
+            np.asarray(self.policy.obstat.mean, dtype=np.float32)).to(self.device))
+        self.obstd.copy_(torch.from_numpy(
+            np.asarray(self.policy.obstat.std, dtype=np.float32)).to(self.device))
+
+    def update_obstat(self, gen_obstat: ObStat):
+        """Fold the generation's stats into the lifetime stat (reference
+        ``policy.update_obstat``, ``policy.py:69-71``) and push to device."""
+        self.policy.update_obstat(gen_obstat)
+        self._push_obstat()
+
+    def sync_host(self, light: bool = False):
+        """Mirror device truth into the host Policy (checkpoint compatibility).
+
+        light=True syncs flat_params only (once per generation); full sync
+        also writes optimizer moments so ``Policy.save`` emits the exact
+        reference pickle format with optimizer state included.
+        """
+        flat = torch.empty(self.n, dtype=torch.float32, device=self.device)
+        flat[self.perm] = self.theta
+        self.policy.flat_params[:] = flat.cpu().numpy()
+        if not light:
+            opt = self.policy.optim
+            inv = torch.empty_like(flat)
+            if isinstance(opt, Adam):
+                inv[self.perm] = self.m
+                opt.m[:] = inv.cpu().numpy()
+                inv[self.perm] = self.v
+                opt.v[:] = inv.cpu().numpy()
+            elif isinstance(opt, SGD):
+                inv[self.perm] = self.v
+                opt.v[:] = inv.cpu().numpy()
+            self.policy.set_nn_params(self.policy.flat_params)

@@ -67,7 +67,8 @@ class BatchedCartPole(BatchedEnv):
         x_dot = x_dot + self.tau * xacc
         theta = theta + self.tau * theta_dot
         theta_dot = theta_dot + self.tau * thetaacc
-        self.state = torch.stack([x, x_dot, theta, theta_dot], dim=1)
+        # in-place into the persistent buffer: hipGraph-capture-safe
+        self.state.copy_(torch.stack([x, x_dot, theta, theta_dot], dim=1))
         self._steps += 1
 
         done = (x.abs() > self.x_threshold) | (theta.abs() > self.theta_threshold) | \
@@ -103,8 +104,8 @@ class BatchedPendulum(BatchedEnv):
         g = torch.Generator(device="cpu")
         if seed is not None:
             g.manual_seed(int(seed))
-        self.th = (torch.rand(self.batch, generator=g) * 2 * math.pi - math.pi).to(self.device)
-        self.thdot = (torch.rand(self.batch, generator=g) * 2 - 1).to(self.device)
+        self.th.copy_((torch.rand(self.batch, generator=g) * 2 * math.pi - math.pi).to(self.device))
+        self.thdot.copy_((torch.rand(self.batch, generator=g) * 2 - 1).to(self.device))
         self._steps.zero_()
         return self._obs()
 
@@ -120,8 +121,8 @@ class BatchedPendulum(BatchedEnv):
         newthdot = self.thdot + (3 * self.g / (2 * self.length) * torch.sin(self.th) +
                                  3.0 / (self.m * self.length ** 2) * u) * self.dt
         newthdot = newthdot.clamp(-self.max_speed, self.max_speed)
-        self.th = self.th + newthdot * self.dt
-        self.thdot = newthdot
+        self.th.add_(newthdot * self.dt)
+        self.thdot.copy_(newthdot)
         self._steps += 1
         done = self._steps >= self.MAX_STEPS
         return self._obs(), -cost, done

@@ -81,13 +81,14 @@ class SyntheticLocomotion(BatchedEnv):
         g = torch.Generator(device="cpu")
         if seed is not None:
             g.manual_seed(int(seed))
-        self.s = (torch.randn(self.batch, self.sdim, generator=g) * 0.1).to(self.device)
+        # in-place into persistent buffers: hipGraph-capture-safe
+        self.s.copy_((torch.randn(self.batch, self.sdim, generator=g) * 0.1).to(self.device))
         self.pos.zero_()
         self._steps.zero_()
         if self.goal_conditioned:
             ang = (torch.rand(self.batch, generator=g) * 2 * np.pi).to(self.device)
             r = 15.0
-            self.goal = torch.stack([r * torch.cos(ang), r * torch.sin(ang)], dim=1)
+            self.goal.copy_(torch.stack([r * torch.cos(ang), r * torch.sin(ang)], dim=1))
         return self._obs()
 
     def _obs(self) -> torch.Tensor:
@@ -99,7 +100,7 @@ class SyntheticLocomotion(BatchedEnv):
     def step(self, actions: torch.Tensor) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor]:
         a = actions.to(self.device).reshape(self.batch, -1).clamp(-1.0, 1.0)
         pre = self.s @ self.A + a @ self.B + self.b0
-        self.s = (1 - self.leak) * self.s + self.leak * torch.tanh(pre)
+        self.s.copy_((1 - self.leak) * self.s + self.leak * torch.tanh(pre))
 
         vfwd = self.s @ self.wv + 0.5 * (a @ self.wa)
         vy = self.s @ self.wy
@@ -114,8 +115,8 @@ class SyntheticLocomotion(BatchedEnv):
         else:
             rew = vfwd - self.ctrl_cost * (a * a).sum(1) + self.alive_bonus
 
-        self.pos = self.pos + self.dt * torch.stack(
-            [vfwd, vy, torch.zeros_like(vfwd)], dim=1)
+        self.pos[:, 0] += self.dt * vfwd
+        self.pos[:, 1] += self.dt * vy
         self.pos[:, 2] = h
         self._steps += 1
 

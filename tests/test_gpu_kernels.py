@@ -1,0 +1,226 @@
+"""HIP kernel numerics vs plain PyTorch fp32 references (run on MI355X).
+
+Each HIP op (noise fill, pheno, fused MLP forward, gather-GEMV, fused Adam)
+is compared against an independent fp32 torch implementation of the same op.
+The reference repo never needed these (single CPU backend) — SURVEY.md §4.
+"""
+import numpy as np
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+
+@pytest.fixture(scope="module")
+def dev():
+    return torch.device("cuda", 0)
+
+
+def _stream(dev):
+    return torch.cuda.current_stream(dev).cuda_stream
+
+
+def test_noise_fill_matches_cpu(dev):
+    from es_pytorch_amd.core.noisetable import NoiseTable
+    n = 1_000_003
+    gpu = NoiseTable.make_noise(n, seed=99, device=dev).cpu()
+    cpu = NoiseTable.make_noise(n, seed=99)
+    # same Philox stream; transcendental libm/ocml ULP differences only
+    assert torch.allclose(gpu, cpu, atol=2e-5, rtol=1e-5)
+    assert (gpu != cpu).float().mean().item() < 0.01  # overwhelmingly bitwise-equal
+
+
+def test_pheno_kernel(dev):
+    from es_pytorch_amd import ops
+    P, n = 7, 1000
+    theta = torch.randn(n, device=dev)
+    table = torch.randn(5000, device=dev)
+    offs = torch.tensor([0, 10, 500, 999, 1234, 4000, 0], dtype=torch.int64, device=dev)
+    signs = torch.tensor([1, 1, 1, -1, -1, -1, 0], dtype=torch.float32, device=dev)
+    out = torch.empty((P, n), dtype=torch.bfloat16, device=dev)
+    std = 0.05
+    ops.check(ops.hip().es_pheno_bf16(out.data_ptr(), theta.data_ptr(), table.data_ptr(),
+                                      offs.data_ptr(), signs.data_ptr(), P, n, std,
+                                      _stream(dev)), "pheno")
+    torch.cuda.synchronize()
+    for p in range(P):
+        expect = (theta + signs[p] * std * table[offs[p]:offs[p] + n]).bfloat16()
+        assert torch.equal(out[p], expect), f"member {p} mismatch"
+    # sign-0 slot is exactly bf16(theta)
+    assert torch.equal(out[6], theta.bfloat16())
+
+
+def _torch_mlp_ref(obs, weights_bf16, dims, obmean, obstd, ob_clip):
+    """fp32 forward with bf16-rounded weights, per the blob layout (W^T, b)."""
+    B = obs.shape[0]
+    x = torch.clamp((obs - obmean) / obstd, -ob_clip, ob_clip)
+    out = []
+    for b in range(B):
+        h = x[b]
+        off = 0
+        for I, O in zip(dims[:-1], dims[1:]):
+            Wt = weights_bf16[b, off:off + I * O].float().reshape(I, O)
+            off += I * O
+            bias = weights_bf16[b, off:off + O].float()
+            off += O
+            h = torch.tanh(h @ Wt + bias)
+        out.append(h)
+    return torch.stack(out)
+
+
+def test_mlp_fwd_kernel(dev):
+    from es_pytorch_amd import ops
+    torch.manual_seed(0)
+    dims = [11, 64, 64, 3]
+    n = sum(I * O + O for I, O in zip(dims[:-1], dims[1:]))
+    B = 9
+    weights = (torch.randn(B, n, device=dev) * 0.3).bfloat16()
+    obs = torch.randn(B, dims[0], device=dev)
+    obmean = torch.randn(dims[0], device=dev) * 0.1
+    obstd = torch.rand(dims[0], device=dev) + 0.5
+    actions = torch.empty(B, dims[-1], device=dev)
+    dims_arr = np.array(dims, dtype=np.int32)
+    ops.check(ops.hip().es_mlp_fwd(actions.data_ptr(), obs.data_ptr(), weights.data_ptr(),
+                                   obmean.data_ptr(), obstd.data_ptr(),
+                                   dims_arr.ctypes.data, len(dims), None, 0, B,
+                                   5.0, 0.0, n, 1, _stream(dev)), "mlp_fwd")
+    torch.cuda.synchronize()
+    ref = _torch_mlp_ref(obs, weights, dims, obmean, obstd, 5.0)
+    assert torch.allclose(actions, ref, atol=2e-2, rtol=2e-2), \
+        (actions - ref).abs().max().item()
+
+
+def test_mlp_fwd_odd_output_dim(dev):
+    """dims with odd final layer exercise the scalar path."""
+    from es_pytorch_amd import ops
+    torch.manual_seed(1)
+    dims = [376, 256, 256, 17]
+    n = sum(I * O + O for I, O in zip(dims[:-1], dims[1:]))
+    B = 3
+    weights = (torch.randn(B, n, device=dev) * 0.1).bfloat16()
+    obs = torch.randn(B, dims[0], device=dev)
+    obmean = torch.zeros(dims[0], device=dev)
+    obstd = torch.ones(dims[0], device=dev)
+    actions = torch.empty(B, dims[-1], device=dev)
+    dims_arr = np.array(dims, dtype=np.int32)
+    ops.check(ops.hip().es_mlp_fwd(actions.data_ptr(), obs.data_ptr(), weights.data_ptr(),
+                                   obmean.data_ptr(), obstd.data_ptr(),
+                                   dims_arr.ctypes.data, len(dims), None, 0, B,
+                                   5.0, 0.0, n, 1, _stream(dev)), "mlp_fwd")
+    torch.cuda.synchronize()
+    ref = _torch_mlp_ref(obs, weights, dims, obmean, obstd, 5.0)
+    assert torch.allclose(actions, ref, atol=2e-2, rtol=2e-2)
+
+
+def test_mlp_fwd_action_noise_statistics(dev):
+    """ac_std noise: mean 0, std ac_std, deterministic in (seed, salt)."""
+    from es_pytorch_amd import ops
+    dims = [4, 8, 2]
+    n = sum(I * O + O for I, O in zip(dims[:-1], dims[1:]))
+    B = 4096
+    weights = torch.zeros(B, n, dtype=torch.bfloat16, device=dev)
+    obs = torch.zeros(B, 4, device=dev)
+    obmean = torch.zeros(4, device=dev)
+    obstd = torch.ones(4, device=dev)
+    a1 = torch.empty(B, 2, device=dev)
+    a2 = torch.empty(B, 2, device=dev)
+    dims_arr = np.array(dims, dtype=np.int32)
+    seed = torch.tensor([123], dtype=torch.int64, device=dev)
+    for out, salt in ((a1, 5), (a2, 5)):
+        ops.check(ops.hip().es_mlp_fwd(out.data_ptr(), obs.data_ptr(), weights.data_ptr(),
+                                       obmean.data_ptr(), obstd.data_ptr(),
+                                       dims_arr.ctypes.data, len(dims), seed.data_ptr(),
+                                       salt, B, 5.0, 0.5, n, 1, _stream(dev)), "mlp_fwd")
+    torch.cuda.synchronize()
+    assert torch.equal(a1, a2)  # same (seed, salt) -> same noise
+    noise = a1.flatten()
+    assert abs(noise.mean().item()) < 0.02
+    assert abs(noise.std().item() - 0.5) < 0.02
+    a3 = torch.empty(B, 2, device=dev)
+    ops.check(ops.hip().es_mlp_fwd(a3.data_ptr(), obs.data_ptr(), weights.data_ptr(),
+                                   obmean.data_ptr(), obstd.data_ptr(),
+                                   dims_arr.ctypes.data, len(dims), seed.data_ptr(),
+                                   6, B, 5.0, 0.5, n, 1, _stream(dev)), "mlp_fwd")
+    torch.cuda.synchronize()
+    assert not torch.equal(a1, a3)  # different salt -> different noise
+
+
+def test_grad_gather_kernel(dev):
+    from es_pytorch_amd import ops
+    torch.manual_seed(2)
+    table = torch.randn(100_000, device=dev)
+    P, n = 333, 4097
+    offs = torch.randint(0, 100_000 - n, (P,), dtype=torch.int64, device=dev)
+    fits = torch.randn(P, device=dev)
+    g = torch.empty(n, device=dev)
+    ops.check(ops.hip().es_grad_gather(g.data_ptr(), table.data_ptr(), fits.data_ptr(),
+                                       offs.data_ptr(), P, n, _stream(dev)), "grad")
+    torch.cuda.synchronize()
+    rows = torch.stack([table[o:o + n] for o in offs.cpu()])
+    expect = fits.cpu() @ rows.cpu()
+    assert torch.allclose(g.cpu(), expect, atol=1e-3, rtol=1e-4)
+
+
+def test_adam_kernel_matches_numpy(dev):
+    from es_pytorch_amd import ops
+    from es_pytorch_amd.nn.optimizers import Adam
+    n = 10_000
+    rng = np.random.RandomState(0)
+    theta0 = rng.randn(n).astype(np.float32)
+    g_np = rng.randn(n).astype(np.float32)
+    l2, n_ranked, lr = 0.005, 64.0, 0.01
+
+    theta = torch.from_numpy(theta0.copy()).to(dev)
+    m = torch.zeros(n, device=dev)
+    v = torch.zeros(n, device=dev)
+    g = torch.from_numpy(g_np * n_ranked).to(dev)  # kernel applies gscale
+    ref_opt = Adam(n, lr)
+    ref_theta = theta0.copy()
+    for t in range(1, 4):
+        a = lr * np.sqrt(1 - ref_opt.beta2 ** t) / (1 - ref_opt.beta1 ** t)
+        ops.check(ops.hip().es_adam_step(theta.data_ptr(), m.data_ptr(), v.data_ptr(),
+                                         g.data_ptr(), n, float(a), 0.9, 0.999, 1e-8,
+                                         l2, 1.0 / n_ranked, _stream(dev)), "adam")
+        # reference semantics: theta += step(l2*theta - grad)
+        ref_theta += ref_opt.step(l2 * ref_theta - g_np)
+    torch.cuda.synchronize()
+    np.testing.assert_allclose(theta.cpu().numpy(), ref_theta, atol=1e-5, rtol=1e-5)
+
+
+def test_engine_graph_matches_eager(dev):
+    """hipGraph-captured rollout == eager rollout for identical seeds."""
+    fits = []
+    for use_graph in (False, True):
+        import numpy as np
+        from es_pytorch_amd.config import AttrDict
+        from es_pytorch_amd.core.engine import GpuEngine
+        from es_pytorch_amd.core.noisetable import NoiseTable
+        from es_pytorch_amd.core.policy import Policy
+        from es_pytorch_amd.envs import make_batched
+        from es_pytorch_amd.nn.nn import FeedForward
+        from es_pytorch_amd.nn.optimizers import Adam
+        from es_pytorch_amd.parallel.comm import Comm
+        from es_pytorch_amd.utils.rankers import CenteredRanker
+
+        torch.manual_seed(3)
+        comm = Comm(dev)
+        cfg = AttrDict({"env": {"name": "Hopper-v3", "max_steps": 40},
+                        "noise": {"tbl_size": 1_000_000, "std": 0.02},
+                        "policy": {"layer_sizes": [64, 64], "ac_std": 0.0, "l2coeff": 0.005,
+                                   "lr": 0.01, "ob_clip": 5, "save_obs_chance": 1.0},
+                        "general": {"policies_per_gen": 8, "batch_size": 500, "seed": 1}})
+        env = make_batched("Hopper-v3", 9, dev, max_steps=40, terminate_on_fall=False)
+        nn = FeedForward([64, 64], torch.nn.Tanh(), env, 0.0, 5)
+        policy = Policy(nn, 0.02, Adam(len(Policy.get_flat(nn)), 0.01))
+        nt = NoiseTable.create_shared(comm, 1_000_000, len(policy), seed=5, device=dev)
+        rs = np.random.RandomState(11)
+        eng = GpuEngine(cfg, comm, policy, nt, env, rs, use_graph=use_graph)
+        ranker = CenteredRanker()
+        eng.step(ranker)
+        fits.append(np.concatenate([ranker.fits_pos, ranker.fits_neg]).ravel())
+    np.testing.assert_allclose(fits[0], fits[1], rtol=1e-5, atol=1e-5)
+
+
+def test_engine_smoke_and_param_motion(dev):
+    import __graft_entry__
+    __graft_entry__.smoke()
