@@ -76,7 +76,8 @@ class _NoiselessResult(RewardResult):
 class GpuEngine:
     def __init__(self, cfg, comm: Comm, policy: Policy, nt: NoiseTable, env: BatchedEnv,
                  rs: np.random.RandomState, objective: str = "reward",
-                 use_graph: bool = True, novelty_k: int = 10):
+                 use_graph: bool = True, novelty_k: int = 10,
+                 fused: Optional[bool] = None):
         self.cfg = cfg
         self.comm = comm
         self.policy = policy
@@ -109,7 +110,10 @@ class GpuEngine:
         self.v = torch.zeros(self.n, dtype=torch.float32, device=d)
         self._load_optim_state()
 
-        self.weights = torch.empty((self.B, self.n), dtype=torch.bfloat16, device=d)
+        # member blob rows padded to 16 B so the forward kernel's uint4
+        # (8 x bf16) vector loads stay aligned for every member
+        self.row_stride = (self.n + 7) // 8 * 8
+        self.weights = torch.empty((self.B, self.row_stride), dtype=torch.bfloat16, device=d)
         self.offsets = torch.zeros(self.B, dtype=torch.int64, device=d)
         self.signs = torch.cat([torch.ones(self.pairs), -torch.ones(self.pairs),
                                 torch.zeros(1)]).to(d)
@@ -139,6 +143,15 @@ class GpuEngine:
         self._graph: Optional[torch.cuda.CUDAGraph] = None
         self.timings = {}
 
+        # fused single-kernel rollout step for the locomotion envs: policy
+        # forward + dynamics + bookkeeping in one launch per step
+        from es_pytorch_amd.envs.locomotion import SyntheticLocomotion
+        self.fused = (isinstance(env, SyntheticLocomotion) and self.device.type == "cuda"
+                      if fused is None else bool(fused))
+        if self.fused:
+            self.mo_sum = torch.zeros((self.B, D), dtype=torch.float32, device=d)
+            self.mo_sumsq = torch.zeros((self.B, D), dtype=torch.float32, device=d)
+
     # ------------------------------------------------------------------ ops
     def _stream(self):
         return torch.cuda.current_stream(self.device).cuda_stream if \
@@ -148,8 +161,8 @@ class GpuEngine:
         std = float(self.policy.std)
         ops.check(ops.hip().es_pheno_bf16(
             self.weights.data_ptr(), self.theta.data_ptr(), self.nt.noise.data_ptr(),
-            self.offsets.data_ptr(), self.signs.data_ptr(), self.B, self.n, std,
-            self._stream()), "es_pheno_bf16")
+            self.offsets.data_ptr(), self.signs.data_ptr(), self.B, self.n,
+            self.row_stride, std, self._stream()), "es_pheno_bf16")
 
     def _forward(self, obs: torch.Tensor, salt: int):
         ac_std = float(getattr(self.policy._module, "_action_std", 0.0))
@@ -157,9 +170,29 @@ class GpuEngine:
             self.actions.data_ptr(), obs.data_ptr(), self.weights.data_ptr(),
             self.obmean.data_ptr(), self.obstd.data_ptr(),
             self.dims_arr.ctypes.data, len(self.dims_arr), self.seed_dev.data_ptr(),
-            salt, self.B, float(self.policy._module.ob_clip), ac_std, self.n, 1,
-            self._stream()), "es_mlp_fwd")
+            salt, self.B, float(self.policy._module.ob_clip), ac_std, self.row_stride, 1,
+            self.B - 1, self._stream()), "es_mlp_fwd")
         return self.actions
+
+    def _loco_step(self, t: int):
+        """One fused rollout step (rollout_loco.hip): forward + env + bookkeeping."""
+        env = self.env
+        ac_std = float(getattr(self.policy._module, "_action_std", 0.0))
+        goal_ptr = env.goal.data_ptr() if env.goal_conditioned else None
+        ops.check(ops.hip().es_loco_step(
+            self.weights.data_ptr(), self.obmean.data_ptr(), self.obstd.data_ptr(),
+            self.dims_arr.ctypes.data, len(self.dims_arr), self.seed_dev.data_ptr(),
+            t + 1, float(self.policy._module.ob_clip), ac_std, self.row_stride,
+            env.s.data_ptr(), env.pos.data_ptr(), goal_ptr,
+            env.A.data_ptr(), env.B.data_ptr(), env.b0.data_ptr(),
+            env.wv.data_ptr(), env.wa.data_ptr(), env.wy.data_ptr(), env.wh.data_ptr(),
+            self.alive.data_ptr(), self.rew_total.data_ptr(),
+            self.member_steps.data_ptr(), self.behv.data_ptr(),
+            self.mo_sum.data_ptr(), self.mo_sumsq.data_ptr(),
+            self.B, env.sdim, env.ac_dim, int(env.goal_conditioned),
+            int(env.terminate_on_fall), self.B - 1,
+            float(env.leak), float(env.ctrl_cost), float(env.alive_bonus),
+            float(env.fall_threshold), float(env.dt), self._stream()), "es_loco_step")
 
     # ------------------------------------------------------------- rollout
     def _step_body(self, t: int):
@@ -181,6 +214,7 @@ class GpuEngine:
         self.obs_buf.copy_(ob)
 
     def _rollout(self):
+        body = self._loco_step if self.fused else self._step_body
         if self.use_graph:
             if self._graph is None:
                 # warmup (lazy inits must happen outside capture), then capture once
@@ -188,7 +222,7 @@ class GpuEngine:
                 s.wait_stream(torch.cuda.current_stream(self.device))
                 with torch.cuda.stream(s):
                     for t in range(3):
-                        self._step_body(t)
+                        body(t)
                 torch.cuda.current_stream(self.device).wait_stream(s)
                 # warmup dirtied the rollout state: restore it before capture
                 self._reset_rollout_state()
@@ -196,11 +230,11 @@ class GpuEngine:
                 self._graph = torch.cuda.CUDAGraph()
                 with torch.cuda.graph(self._graph):
                     for t in range(self.max_steps):
-                        self._step_body(t)
+                        body(t)
             self._graph.replay()
             return
         for t in range(self.max_steps):
-            self._step_body(t)
+            body(t)
 
     def _reset_rollout_state(self):
         self.alive.fill_(1.0)
@@ -211,6 +245,9 @@ class GpuEngine:
         self.ob_sum.zero_()
         self.ob_sumsq.zero_()
         self.ob_count.zero_()
+        if self.fused:
+            self.mo_sum.zero_()
+            self.mo_sumsq.zero_()
 
     def _gen_seed(self) -> int:
         # distinct env variations per generation, identical across ranks only
@@ -267,7 +304,10 @@ class GpuEngine:
             torch.cuda.synchronize(self.device)
         t2 = time.perf_counter()
         all_rows = self.comm.allgather_rows(rows).cpu().numpy()
-        steps = int(self.comm.allreduce_scalar(float(self.steps_total.item())))
+        # evaluated-episode steps only (the reference's es.py:79 counts the
+        # pos+neg rollouts, not the noiseless eval)
+        local_steps = float(self.member_steps[:2 * self.pairs].sum().item())
+        steps = int(self.comm.allreduce_scalar(local_steps))
         t3 = time.perf_counter()
 
         # -- identical redundant ranking on every rank (host; pop-sized)
@@ -289,8 +329,17 @@ class GpuEngine:
 
         # -- per-gen obstat -> merged across ranks (packed all_reduce)
         gen_obstat = ObStat(self.env.observation_space.shape, 0)
-        gen_obstat.inc(self.ob_sum.cpu().numpy(), self.ob_sumsq.cpu().numpy(),
-                       float(self.ob_count.item()))
+        if self.fused:
+            # fused kernel accumulated per-member alive-weighted sums; apply
+            # the save_obs_chance member flags at reduction time
+            sm = self.save_mask.unsqueeze(1)
+            ob_sum = (self.mo_sum * sm).sum(0).double().cpu().numpy()
+            ob_sumsq = (self.mo_sumsq * sm).sum(0).double().cpu().numpy()
+            ob_count = float((self.member_steps * self.save_mask).sum().item())
+            gen_obstat.inc(ob_sum, ob_sumsq, ob_count)
+        else:
+            gen_obstat.inc(self.ob_sum.cpu().numpy(), self.ob_sumsq.cpu().numpy(),
+                           float(self.ob_count.item()))
         gen_obstat.dist_inc(self.comm)
 
         nl_rew = float(self.rew_total[-1].item())

@@ -1,0 +1,109 @@
+// Shared device-side MLP forward core, used by the standalone forward kernel
+// (mlp_fwd.hip) and the fused rollout-step kernel (rollout_loco.hip).
+//
+// Scheme (HBM-bandwidth-driven, guide §2/G13): per layer, thread t owns
+// output octet oi = t % (O/8) and i-partition ip = t / (O/8); per i it loads
+// W^T[i][8*oi..+7] as one 16-B uint4 (1 KiB per wave-instruction) and FMAs
+// into 8 independent accumulator chains; partials reduce across i-partitions
+// through LDS; bias + tanh in the epilogue. Non-8-aligned layers (the tiny
+// action head) take a scalar fallback.
+#pragma once
+#include "common.h"
+
+#define ES_MAXL 8
+#define ES_MAXDIM 2048
+
+struct MlpShape {
+  int n_layers;
+  int dims[ES_MAXL + 1];
+  int64_t woff[ES_MAXL];
+  int64_t boff[ES_MAXL];
+  int vec_ok[ES_MAXL];
+};
+
+// Host-side shape builder; returns 0 on success.
+static inline int mlp_shape_init(MlpShape* sh, const int32_t* dims_host, int32_t ndims,
+                                 int64_t row_stride) {
+  if (ndims < 2 || ndims > ES_MAXL + 1) return -100;
+  sh->n_layers = ndims - 1;
+  int64_t off = 0;
+  for (int l = 0; l < ndims; ++l) {
+    sh->dims[l] = dims_host[l];
+    if (dims_host[l] > ES_MAXDIM) return -101;
+  }
+  for (int l = 0; l < sh->n_layers; ++l) {
+    sh->woff[l] = off;
+    off += (int64_t)sh->dims[l] * sh->dims[l + 1];
+    sh->boff[l] = off;
+    off += sh->dims[l + 1];
+    const int O = sh->dims[l + 1];
+    sh->vec_ok[l] = (O % 8 == 0) && (O <= 8 * 256) && (sh->woff[l] % 8 == 0);
+  }
+  if (off > row_stride) return -102;
+  return 0;
+}
+
+__device__ __forceinline__ void bf8_fma(uint4 w, const float xi, float* acc) {
+  const uint32_t ws[4] = {w.x, w.y, w.z, w.w};
+#pragma unroll
+  for (int q = 0; q < 4; ++q) {
+    acc[2 * q] = fmaf(bf2f((uint16_t)(ws[q] & 0xFFFFu)), xi, acc[2 * q]);
+    acc[2 * q + 1] = fmaf(bf2f((uint16_t)(ws[q] >> 16)), xi, acc[2 * q + 1]);
+  }
+}
+
+// Runs every layer; input in buf[0]; returns the index of the buffer holding
+// the final output. `partial` is a 256*8-float LDS scratch.
+__device__ __forceinline__ int mlp_layers(const uint16_t* __restrict__ wb,
+                                          const MlpShape& sh,
+                                          float (*buf)[ES_MAXDIM], float* partial,
+                                          int tid, int nthreads, int act_final) {
+  int cur = 0;
+  for (int l = 0; l < sh.n_layers; ++l) {
+    const int I = sh.dims[l], O = sh.dims[l + 1];
+    const uint16_t* Wt = wb + sh.woff[l];
+    const uint16_t* Bs = wb + sh.boff[l];
+    const bool do_act = (l < sh.n_layers - 1) || act_final;
+    const float* x = buf[cur];
+    float* y = buf[cur ^ 1];
+
+    if (sh.vec_ok[l]) {
+      const int OCT = O >> 3;
+      const int PART = nthreads / OCT;
+      const int oi = tid % OCT, ip = tid / OCT;
+      float acc[8];
+#pragma unroll
+      for (int q = 0; q < 8; ++q) acc[q] = 0.0f;
+      if (ip < PART) {
+        const uint16_t* wcol = Wt + (oi << 3);
+        for (int i = ip; i < I; i += PART) {
+          const uint4 w = *reinterpret_cast<const uint4*>(wcol + (int64_t)i * O);
+          bf8_fma(w, x[i], acc);
+        }
+#pragma unroll
+        for (int q = 0; q < 8; ++q) partial[(ip * OCT + oi) * 8 + q] = acc[q];
+      }
+      __syncthreads();
+      for (int o = tid; o < O; o += nthreads) {
+        float s = bf2f(Bs[o]);
+        const int oo = o >> 3, j = o & 7;
+        for (int p = 0; p < PART; ++p) s += partial[(p * OCT + oo) * 8 + j];
+        y[o] = do_act ? tanhf(s) : s;
+      }
+    } else {
+      for (int o = tid; o < O; o += nthreads) {
+        float acc = bf2f(Bs[o]);
+        for (int i = 0; i < I; ++i) acc = fmaf(bf2f(Wt[(int64_t)i * O + o]), x[i], acc);
+        y[o] = do_act ? tanhf(acc) : acc;
+      }
+    }
+    __syncthreads();
+    cur ^= 1;
+  }
+  return cur;
+}
+
+__device__ __forceinline__ float es_actnoise(uint64_t seed, uint64_t ctr) {
+  esrng::f32x4 v = esrng::normal4(ctr, seed, 0xACu);
+  return v.x;
+}

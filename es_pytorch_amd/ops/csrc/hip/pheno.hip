@@ -14,11 +14,11 @@ __global__ void pheno_bf16_kernel(uint16_t* __restrict__ out, const float* __res
                                   const float* __restrict__ table,
                                   const int64_t* __restrict__ offsets,
                                   const float* __restrict__ signs, int64_t n_params,
-                                  float std) {
+                                  int64_t row_stride, float std) {
   int64_t b = blockIdx.y;
   const float s = signs[b] * std;
   const float* noise = table + offsets[b];
-  uint16_t* ob = out + b * n_params;
+  uint16_t* ob = out + b * row_stride;
   for (int64_t t = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; t < n_params;
        t += (int64_t)gridDim.x * blockDim.x) {
     ob[t] = f2bf(theta[t] + s * noise[t]);
@@ -27,13 +27,13 @@ __global__ void pheno_bf16_kernel(uint16_t* __restrict__ out, const float* __res
 
 extern "C" int es_pheno_bf16(void* out, const void* theta, const void* table,
                              const void* offsets, const void* signs, int64_t n_pop,
-                             int64_t n_params, float std, void* stream) {
+                             int64_t n_params, int64_t row_stride, float std, void* stream) {
   int threads = 256;
   int bx = (int)std::min<int64_t>((n_params + threads - 1) / threads, 1024);
   dim3 grid(bx, (unsigned)n_pop);
   pheno_bf16_kernel<<<grid, dim3(threads), 0, (hipStream_t)stream>>>(
       (uint16_t*)out, (const float*)theta, (const float*)table, (const int64_t*)offsets,
-      (const float*)signs, n_params, std);
+      (const float*)signs, n_params, row_stride, std);
   ES_CHECK_LAUNCH();
   return 0;
 }
