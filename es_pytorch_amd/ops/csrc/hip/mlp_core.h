@@ -19,7 +19,13 @@ struct MlpShape {
   int64_t woff[ES_MAXL];
   int64_t boff[ES_MAXL];
   int vec_ok[ES_MAXL];
+  int maxdim;  // max layer width, 4-padded — LDS activation buffer extent
 };
+
+// dynamic-LDS carve sizes (bytes, 16-B aligned each; guide §6 G17)
+static inline int64_t mlp_lds_bytes(int maxdim) {
+  return 2 * (int64_t)maxdim * 4 + 256 * 8 * 4;  // buf[2][maxdim] + partial
+}
 
 // Host-side shape builder; returns 0 on success.
 static inline int mlp_shape_init(MlpShape* sh, const int32_t* dims_host, int32_t ndims,
@@ -27,10 +33,13 @@ static inline int mlp_shape_init(MlpShape* sh, const int32_t* dims_host, int32_t
   if (ndims < 2 || ndims > ES_MAXL + 1) return -100;
   sh->n_layers = ndims - 1;
   int64_t off = 0;
+  sh->maxdim = 0;
   for (int l = 0; l < ndims; ++l) {
     sh->dims[l] = dims_host[l];
     if (dims_host[l] > ES_MAXDIM) return -101;
+    if (dims_host[l] > sh->maxdim) sh->maxdim = dims_host[l];
   }
+  sh->maxdim = (sh->maxdim + 3) & ~3;
   for (int l = 0; l < sh->n_layers; ++l) {
     sh->woff[l] = off;
     off += (int64_t)sh->dims[l] * sh->dims[l + 1];
@@ -52,20 +61,19 @@ __device__ __forceinline__ void bf8_fma(uint4 w, const float xi, float* acc) {
   }
 }
 
-// Runs every layer; input in buf[0]; returns the index of the buffer holding
-// the final output. `partial` is a 256*8-float LDS scratch.
-__device__ __forceinline__ int mlp_layers(const uint16_t* __restrict__ wb,
-                                          const MlpShape& sh,
-                                          float (*buf)[ES_MAXDIM], float* partial,
-                                          int tid, int nthreads, int act_final) {
-  int cur = 0;
+// Runs every layer; input in bufA; returns the pointer holding the final
+// output (bufA or bufB). `partial` is a 256*8-float LDS scratch.
+__device__ __forceinline__ float* mlp_layers(const uint16_t* __restrict__ wb,
+                                             const MlpShape& sh,
+                                             float* bufA, float* bufB, float* partial,
+                                             int tid, int nthreads, int act_final) {
+  float* x = bufA;
+  float* y = bufB;
   for (int l = 0; l < sh.n_layers; ++l) {
     const int I = sh.dims[l], O = sh.dims[l + 1];
     const uint16_t* Wt = wb + sh.woff[l];
     const uint16_t* Bs = wb + sh.boff[l];
     const bool do_act = (l < sh.n_layers - 1) || act_final;
-    const float* x = buf[cur];
-    float* y = buf[cur ^ 1];
 
     if (sh.vec_ok[l]) {
       const int OCT = O >> 3;
@@ -106,16 +114,62 @@ __device__ __forceinline__ int mlp_layers(const uint16_t* __restrict__ wb,
         y[o] = do_act ? tanhf(s) : s;
       }
     } else {
-      for (int o = tid; o < O; o += nthreads) {
-        float acc = bf2f(Bs[o]);
-        for (int i = 0; i < I; ++i) acc = fmaf(bf2f(Wt[(int64_t)i * O + o]), x[i], acc);
-        y[o] = do_act ? tanhf(acc) : acc;
+      // tiled scalar path (small / odd layers, e.g. the 256->17 action head):
+      // thread owns (output o, i-partition ip) so the whole block shares the
+      // I-dim walk — a thread-per-output loop leaves O threads issuing I
+      // serial HBM-latency loads each (measured: the action head dominated
+      // the fused step kernel that way).
+      const int PART = nthreads / O;
+      if (PART > 1) {
+        const int oi = tid % O, ip = tid / O;
+        float acc = 0.0f;
+        if (ip < PART) {
+          int i = ip;
+          const int step4 = PART * 4;
+          for (; i + 3 * PART < I; i += step4) {
+            const float w0 = bf2f(Wt[(int64_t)i * O + oi]);
+            const float w1 = bf2f(Wt[(int64_t)(i + PART) * O + oi]);
+            const float w2 = bf2f(Wt[(int64_t)(i + 2 * PART) * O + oi]);
+            const float w3 = bf2f(Wt[(int64_t)(i + 3 * PART) * O + oi]);
+            acc = fmaf(w0, x[i], acc);
+            acc = fmaf(w1, x[i + PART], acc);
+            acc = fmaf(w2, x[i + 2 * PART], acc);
+            acc = fmaf(w3, x[i + 3 * PART], acc);
+          }
+          for (; i < I; i += PART) acc = fmaf(bf2f(Wt[(int64_t)i * O + oi]), x[i], acc);
+          partial[ip * O + oi] = acc;
+        }
+        __syncthreads();
+        for (int o = tid; o < O; o += nthreads) {
+          float s = bf2f(Bs[o]);
+          for (int p = 0; p < PART; ++p) s += partial[p * O + o];
+          y[o] = do_act ? tanhf(s) : s;
+        }
+      } else {
+        for (int o = tid; o < O; o += nthreads) {
+          float acc = bf2f(Bs[o]);
+          int i = 0;
+          for (; i + 3 < I; i += 4) {
+            const float w0 = bf2f(Wt[(int64_t)i * O + o]);
+            const float w1 = bf2f(Wt[(int64_t)(i + 1) * O + o]);
+            const float w2 = bf2f(Wt[(int64_t)(i + 2) * O + o]);
+            const float w3 = bf2f(Wt[(int64_t)(i + 3) * O + o]);
+            acc = fmaf(w0, x[i], acc);
+            acc = fmaf(w1, x[i + 1], acc);
+            acc = fmaf(w2, x[i + 2], acc);
+            acc = fmaf(w3, x[i + 3], acc);
+          }
+          for (; i < I; ++i) acc = fmaf(bf2f(Wt[(int64_t)i * O + o]), x[i], acc);
+          y[o] = do_act ? tanhf(acc) : acc;
+        }
       }
     }
     __syncthreads();
-    cur ^= 1;
+    float* t = x;
+    x = y;
+    y = t;
   }
-  return cur;
+  return x;
 }
 
 __device__ __forceinline__ float es_actnoise(uint64_t seed, uint64_t ctr) {

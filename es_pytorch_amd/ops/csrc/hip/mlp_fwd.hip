@@ -14,8 +14,10 @@ mlp_fwd_kernel(float* __restrict__ actions, const float* __restrict__ obs,
                const float* __restrict__ obstd, MlpShape sh, float ob_clip, float ac_std,
                const uint64_t* __restrict__ seed_dev, uint64_t salt, int64_t row_stride,
                int act_final, int noiseless_from) {
-  __shared__ float buf[2][ES_MAXDIM];
-  __shared__ float partial[256 * 8];
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* bufA = reinterpret_cast<float*>(smem);
+  float* bufB = bufA + sh.maxdim;
+  float* partial = bufB + sh.maxdim;
   const int b = blockIdx.x;
   const int tid = threadIdx.x;
   const uint16_t* wb = weights + (int64_t)b * row_stride;
@@ -23,16 +25,16 @@ mlp_fwd_kernel(float* __restrict__ actions, const float* __restrict__ obs,
 
   for (int i = tid; i < D; i += blockDim.x) {
     float v = (obs[(int64_t)b * D + i] - obmean[i]) / obstd[i];
-    buf[0][i] = fclampf(v, -ob_clip, ob_clip);
+    bufA[i] = fclampf(v, -ob_clip, ob_clip);
   }
   __syncthreads();
 
-  const int cur = mlp_layers(wb, sh, buf, partial, tid, blockDim.x, act_final);
+  const float* out = mlp_layers(wb, sh, bufA, bufB, partial, tid, blockDim.x, act_final);
 
   const int A = sh.dims[sh.n_layers];
   const uint64_t seed = seed_dev ? (*seed_dev + salt) : salt;
   for (int o = tid; o < A; o += blockDim.x) {
-    float a = buf[cur][o];
+    float a = out[o];
     // members >= noiseless_from are evaluated without action noise
     // (the reference's noiseless eval passes rs=None, es.py:48)
     if (ac_std != 0.0f && b < noiseless_from)
@@ -49,7 +51,8 @@ extern "C" int es_mlp_fwd(void* actions, const void* obs, const void* weights,
   MlpShape sh;
   int rc = mlp_shape_init(&sh, dims_host, ndims, row_stride);
   if (rc) return rc;
-  mlp_fwd_kernel<<<dim3((unsigned)n_pop), dim3(256), 0, (hipStream_t)stream>>>(
+  mlp_fwd_kernel<<<dim3((unsigned)n_pop), dim3(256), (unsigned)mlp_lds_bytes(sh.maxdim),
+                   (hipStream_t)stream>>>(
       (float*)actions, (const float*)obs, (const uint16_t*)weights, (const float*)obmean,
       (const float*)obstd, sh, ob_clip, ac_std, (const uint64_t*)seed_dev, salt, row_stride,
       act_final, noiseless_from);

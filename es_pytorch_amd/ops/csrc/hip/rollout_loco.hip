@@ -39,11 +39,13 @@ loco_step_kernel(const uint16_t* __restrict__ weights, const float* __restrict__
                  float* __restrict__ alive, float* __restrict__ rew_total,
                  float* __restrict__ member_steps, float* __restrict__ behv,
                  float* __restrict__ mo_sum, float* __restrict__ mo_sumsq) {
-  __shared__ float buf[2][ES_MAXDIM];
-  __shared__ float partial[256 * 8];
-  __shared__ float raws[ES_MAXDIM];
-  __shared__ float abuf[64];
-  __shared__ float sc[8];  // 0:vfwd 1:vy 2:h 3:asq 4:alive_old 5:rew 6:done
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* bufA = reinterpret_cast<float*>(smem);
+  float* bufB = bufA + sh.maxdim;
+  float* partial = bufB + sh.maxdim;
+  float* raws = partial + 256 * 8;
+  float* abuf = raws + ((la.S + 3) & ~3);
+  float* sc = abuf + 64;  // [4]: alive_old broadcast
   const int b = blockIdx.x;
   const int tid = threadIdx.x;
   const int nth = blockDim.x;
@@ -54,21 +56,21 @@ loco_step_kernel(const uint16_t* __restrict__ weights, const float* __restrict__
   for (int i = tid; i < S; i += nth) {
     const float v = sb[i];
     raws[i] = v;
-    buf[0][i] = fclampf((v - obmean[i]) / obstd[i], -la.ob_clip, la.ob_clip);
+    bufA[i] = fclampf((v - obmean[i]) / obstd[i], -la.ob_clip, la.ob_clip);
   }
   if (la.goal && tid < 2) {
     const float rel = (goal[(int64_t)b * 2 + tid] - pos[(int64_t)b * 3 + tid]) * 0.1f;
-    buf[0][S + tid] = fclampf((rel - obmean[S + tid]) / obstd[S + tid], -la.ob_clip,
-                              la.ob_clip);
+    bufA[S + tid] = fclampf((rel - obmean[S + tid]) / obstd[S + tid], -la.ob_clip,
+                            la.ob_clip);
   }
   __syncthreads();
 
   // ---- policy forward ----------------------------------------------------
   const uint16_t* wb = weights + (int64_t)b * la.row_stride;
-  const int cur = mlp_layers(wb, sh, buf, partial, tid, nth, 1);
+  const float* aout = mlp_layers(wb, sh, bufA, bufB, partial, tid, nth, 1);
   const uint64_t seed = seed_dev ? (*seed_dev + la.salt) : la.salt;
   if (tid < A) {
-    float a = buf[cur][tid];
+    float a = aout[tid];
     if (la.ac_std != 0.0f && b < la.noiseless_from)
       a += la.ac_std * es_actnoise(seed, (uint64_t)b * A + tid);
     abuf[tid] = fclampf(a, -1.0f, 1.0f);  // env action clamp (locomotion.py)
@@ -118,7 +120,7 @@ loco_step_kernel(const uint16_t* __restrict__ weights, const float* __restrict__
         for (int pp = 0; pp < PART; ++pp) p += partial[(pp * OCT + oo) * 4 + j];
         for (int k = 0; k < A; ++k) p = fmaf(abuf[k], Bm[(int64_t)k * S + o], p);
         const float sn = (1.0f - la.leak) * raws[o] + la.leak * tanhf(p);
-        buf[0][o] = sn;
+        bufA[o] = sn;
         sb[o] = sn;
       }
     } else {
@@ -127,7 +129,7 @@ loco_step_kernel(const uint16_t* __restrict__ weights, const float* __restrict__
         for (int i = 0; i < S; ++i) p = fmaf(raws[i], Am[(int64_t)i * S + o], p);
         for (int k = 0; k < A; ++k) p = fmaf(abuf[k], Bm[(int64_t)k * S + o], p);
         const float sn = (1.0f - la.leak) * raws[o] + la.leak * tanhf(p);
-        buf[0][o] = sn;
+        bufA[o] = sn;
         sb[o] = sn;
       }
     }
@@ -138,7 +140,7 @@ loco_step_kernel(const uint16_t* __restrict__ weights, const float* __restrict__
   {
     float p0 = 0, p1 = 0, p2 = 0, p3 = 0;
     for (int i = tid; i < S; i += nth) {
-      const float sn = buf[0][i];
+      const float sn = bufA[i];
       p0 = fmaf(sn, wv[i], p0);
       p1 = fmaf(sn, wy[i], p1);
       p2 = fmaf(sn, wh[i], p2);
@@ -200,7 +202,7 @@ loco_step_kernel(const uint16_t* __restrict__ weights, const float* __restrict__
     float* ms = mo_sum + (int64_t)b * la.D;
     float* mq = mo_sumsq + (int64_t)b * la.D;
     for (int i = tid; i < S; i += nth) {
-      const float o = buf[0][i];
+      const float o = bufA[i];
       ms[i] += o;
       mq[i] += o * o;
     }
@@ -235,7 +237,9 @@ extern "C" int es_loco_step(const void* weights, const void* obmean, const void*
   la.leak = leak; la.ctrl = ctrl; la.alive_bonus = alive_bonus; la.fall_thr = fall_thr;
   la.dt = dt; la.ob_clip = ob_clip; la.ac_std = ac_std; la.salt = salt;
   la.row_stride = row_stride;
-  loco_step_kernel<<<dim3((unsigned)n_pop), dim3(256), 0, (hipStream_t)stream>>>(
+  const unsigned lds = (unsigned)(mlp_lds_bytes(sh.maxdim) +
+                                  (((sdim + 3) & ~3) + 64 + 8) * 4);
+  loco_step_kernel<<<dim3((unsigned)n_pop), dim3(256), lds, (hipStream_t)stream>>>(
       (const uint16_t*)weights, (const float*)obmean, (const float*)obstd, sh, la,
       (const uint64_t*)seed_dev, (float*)s_glob, (float*)pos, (const float*)goal,
       (const float*)Am, (const float*)Bm, (const float*)b0, (const float*)wv,
