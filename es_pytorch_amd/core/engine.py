@@ -119,6 +119,8 @@ class GpuEngine:
                                 torch.zeros(1)]).to(d)
         self.grad = torch.empty(self.n, dtype=torch.float32, device=d)
         self.seed_dev = torch.zeros(1, dtype=torch.int64, device=d)
+        # device-resident mutable scalars read by kernels (graph-replay-safe)
+        self.acstd_dev = torch.zeros(1, dtype=torch.float32, device=d)
 
         D = env.ob_dim
         self.obmean = torch.zeros(D, dtype=torch.float32, device=d)
@@ -165,24 +167,23 @@ class GpuEngine:
             self.row_stride, std, self._stream()), "es_pheno_bf16")
 
     def _forward(self, obs: torch.Tensor, salt: int):
-        ac_std = float(getattr(self.policy._module, "_action_std", 0.0))
         ops.check(ops.hip().es_mlp_fwd(
             self.actions.data_ptr(), obs.data_ptr(), self.weights.data_ptr(),
             self.obmean.data_ptr(), self.obstd.data_ptr(),
             self.dims_arr.ctypes.data, len(self.dims_arr), self.seed_dev.data_ptr(),
-            salt, self.B, float(self.policy._module.ob_clip), ac_std, self.row_stride, 1,
-            self.B - 1, self._stream()), "es_mlp_fwd")
+            salt, self.B, float(self.policy._module.ob_clip), self.acstd_dev.data_ptr(),
+            self.row_stride, 1, self.B - 1, self._stream()), "es_mlp_fwd")
         return self.actions
 
     def _loco_step(self, t: int):
         """One fused rollout step (rollout_loco.hip): forward + env + bookkeeping."""
         env = self.env
-        ac_std = float(getattr(self.policy._module, "_action_std", 0.0))
         goal_ptr = env.goal.data_ptr() if env.goal_conditioned else None
         ops.check(ops.hip().es_loco_step(
             self.weights.data_ptr(), self.obmean.data_ptr(), self.obstd.data_ptr(),
             self.dims_arr.ctypes.data, len(self.dims_arr), self.seed_dev.data_ptr(),
-            t + 1, float(self.policy._module.ob_clip), ac_std, self.row_stride,
+            t + 1, float(self.policy._module.ob_clip), self.acstd_dev.data_ptr(),
+            self.row_stride,
             env.s.data_ptr(), env.pos.data_ptr(), goal_ptr,
             env.A.data_ptr(), env.B.data_ptr(), env.b0.data_ptr(),
             env.wv.data_ptr(), env.wa.data_ptr(), env.wy.data_ptr(), env.wh.data_ptr(),
@@ -282,6 +283,7 @@ class GpuEngine:
         self.offsets[:self.pairs].copy_(torch.from_numpy(offs).to(self.device))
         self.offsets[self.pairs:2 * self.pairs].copy_(self.offsets[:self.pairs])
         self.seed_dev.fill_(int(self.rs.randint(0, 2 ** 31)))
+        self.acstd_dev.fill_(float(getattr(self.policy._module, "_action_std", 0.0)))
         chance = float(cfg.policy.get("save_obs_chance", 1.0))
         sm = (self.rs.random_sample(self.B) < chance).astype(np.float32)
         sm[-1] = 0.0  # noiseless slot never feeds the gen obstat (reference es.py:48 is separate)

@@ -109,12 +109,12 @@ def _bind_hip(lib):
                                   ctypes.c_int32, ctypes.c_float, ctypes.c_void_p)
     lib.es_noise_fill.argtypes = [p, i64, u64, u32, p]
     lib.es_pheno_bf16.argtypes = [p, p, p, p, p, i64, i64, i64, f32, p]
-    lib.es_mlp_fwd.argtypes = [p, p, p, p, p, p, i32, p, u64, i32, f32, f32, i64, i32,
+    lib.es_mlp_fwd.argtypes = [p, p, p, p, p, p, i32, p, u64, i32, f32, p, i64, i32,
                                i32, p]
     lib.es_grad_gather.argtypes = [p, p, p, p, i64, i64, p]
     lib.es_adam_step.argtypes = [p, p, p, p, i64, f32, f32, f32, f32, f32, f32, p]
     lib.es_sgd_step.argtypes = [p, p, p, i64, f32, f32, f32, f32, p]
-    lib.es_loco_step.argtypes = [p, p, p, p, i32, p, u64, f32, f32, i64,
+    lib.es_loco_step.argtypes = [p, p, p, p, i32, p, u64, f32, p, i64,
                                  p, p, p, p, p, p, p, p, p, p,
                                  p, p, p, p, p, p,
                                  i32, i32, i32, i32, i32, i32,

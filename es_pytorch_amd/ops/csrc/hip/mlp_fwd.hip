@@ -11,7 +11,8 @@
 __global__ void __launch_bounds__(256)
 mlp_fwd_kernel(float* __restrict__ actions, const float* __restrict__ obs,
                const uint16_t* __restrict__ weights, const float* __restrict__ obmean,
-               const float* __restrict__ obstd, MlpShape sh, float ob_clip, float ac_std,
+               const float* __restrict__ obstd, MlpShape sh, float ob_clip,
+               const float* __restrict__ ac_std_dev,
                const uint64_t* __restrict__ seed_dev, uint64_t salt, int64_t row_stride,
                int act_final, int noiseless_from) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
@@ -33,6 +34,9 @@ mlp_fwd_kernel(float* __restrict__ actions, const float* __restrict__ obs,
 
   const int A = sh.dims[sh.n_layers];
   const uint64_t seed = seed_dev ? (*seed_dev + salt) : salt;
+  // ac_std is read from device memory so decay schedules keep working under
+  // hipGraph replay (kernel args are frozen at capture time)
+  const float ac_std = ac_std_dev ? *ac_std_dev : 0.0f;
   for (int o = tid; o < A; o += blockDim.x) {
     float a = out[o];
     // members >= noiseless_from are evaluated without action noise
@@ -46,16 +50,16 @@ mlp_fwd_kernel(float* __restrict__ actions, const float* __restrict__ obs,
 extern "C" int es_mlp_fwd(void* actions, const void* obs, const void* weights,
                           const void* obmean, const void* obstd, const int32_t* dims_host,
                           int32_t ndims, const void* seed_dev, uint64_t salt, int32_t n_pop,
-                          float ob_clip, float ac_std, int64_t row_stride, int32_t act_final,
-                          int32_t noiseless_from, void* stream) {
+                          float ob_clip, const void* ac_std_dev, int64_t row_stride,
+                          int32_t act_final, int32_t noiseless_from, void* stream) {
   MlpShape sh;
   int rc = mlp_shape_init(&sh, dims_host, ndims, row_stride);
   if (rc) return rc;
   mlp_fwd_kernel<<<dim3((unsigned)n_pop), dim3(256), (unsigned)mlp_lds_bytes(sh.maxdim),
                    (hipStream_t)stream>>>(
       (float*)actions, (const float*)obs, (const uint16_t*)weights, (const float*)obmean,
-      (const float*)obstd, sh, ob_clip, ac_std, (const uint64_t*)seed_dev, salt, row_stride,
-      act_final, noiseless_from);
+      (const float*)obstd, sh, ob_clip, (const float*)ac_std_dev,
+      (const uint64_t*)seed_dev, salt, row_stride, act_final, noiseless_from);
   ES_CHECK_LAUNCH();
   return 0;
 }

@@ -22,7 +22,7 @@ struct LocoArgs {
   int goal;         // goal-conditioned flag
   int terminate;    // terminate_on_fall
   int noiseless_from;  // members >= this index get no action noise
-  float leak, ctrl, alive_bonus, fall_thr, dt, ob_clip, ac_std;
+  float leak, ctrl, alive_bonus, fall_thr, dt, ob_clip;
   uint64_t salt;
   int64_t row_stride;
 };
@@ -30,6 +30,7 @@ struct LocoArgs {
 __global__ void __launch_bounds__(256)
 loco_step_kernel(const uint16_t* __restrict__ weights, const float* __restrict__ obmean,
                  const float* __restrict__ obstd, MlpShape sh, LocoArgs la,
+                 const float* __restrict__ ac_std_dev,
                  const uint64_t* __restrict__ seed_dev,
                  float* __restrict__ s_glob, float* __restrict__ pos,
                  const float* __restrict__ goal, const float* __restrict__ Am,
@@ -69,10 +70,11 @@ loco_step_kernel(const uint16_t* __restrict__ weights, const float* __restrict__
   const uint16_t* wb = weights + (int64_t)b * la.row_stride;
   const float* aout = mlp_layers(wb, sh, bufA, bufB, partial, tid, nth, 1);
   const uint64_t seed = seed_dev ? (*seed_dev + la.salt) : la.salt;
+  const float ac_std = ac_std_dev ? *ac_std_dev : 0.0f;  // device-read: graph-safe decay
   if (tid < A) {
     float a = aout[tid];
-    if (la.ac_std != 0.0f && b < la.noiseless_from)
-      a += la.ac_std * es_actnoise(seed, (uint64_t)b * A + tid);
+    if (ac_std != 0.0f && b < la.noiseless_from)
+      a += ac_std * es_actnoise(seed, (uint64_t)b * A + tid);
     abuf[tid] = fclampf(a, -1.0f, 1.0f);  // env action clamp (locomotion.py)
   }
   __syncthreads();
@@ -216,7 +218,8 @@ loco_step_kernel(const uint16_t* __restrict__ weights, const float* __restrict__
 
 extern "C" int es_loco_step(const void* weights, const void* obmean, const void* obstd,
                             const int32_t* dims_host, int32_t ndims, const void* seed_dev,
-                            uint64_t salt, float ob_clip, float ac_std, int64_t row_stride,
+                            uint64_t salt, float ob_clip, const void* ac_std_dev,
+                            int64_t row_stride,
                             void* s_glob, void* pos, const void* goal, const void* Am,
                             const void* Bm, const void* b0, const void* wv, const void* wa,
                             const void* wy, const void* wh, void* alive, void* rew_total,
@@ -235,13 +238,13 @@ extern "C" int es_loco_step(const void* weights, const void* obmean, const void*
   la.S = sdim; la.A = adim; la.D = sh.dims[0]; la.goal = goal_flag;
   la.terminate = terminate; la.noiseless_from = noiseless_from;
   la.leak = leak; la.ctrl = ctrl; la.alive_bonus = alive_bonus; la.fall_thr = fall_thr;
-  la.dt = dt; la.ob_clip = ob_clip; la.ac_std = ac_std; la.salt = salt;
+  la.dt = dt; la.ob_clip = ob_clip; la.salt = salt;
   la.row_stride = row_stride;
   const unsigned lds = (unsigned)(mlp_lds_bytes(sh.maxdim) +
                                   (((sdim + 3) & ~3) + 64 + 8) * 4);
   loco_step_kernel<<<dim3((unsigned)n_pop), dim3(256), lds, (hipStream_t)stream>>>(
       (const uint16_t*)weights, (const float*)obmean, (const float*)obstd, sh, la,
-      (const uint64_t*)seed_dev, (float*)s_glob, (float*)pos, (const float*)goal,
+      (const float*)ac_std_dev, (const uint64_t*)seed_dev, (float*)s_glob, (float*)pos, (const float*)goal,
       (const float*)Am, (const float*)Bm, (const float*)b0, (const float*)wv,
       (const float*)wa, (const float*)wy, (const float*)wh, (float*)alive,
       (float*)rew_total, (float*)member_steps, (float*)behv, (float*)mo_sum,
