@@ -395,6 +395,18 @@ class GpuEngine:
         self.policy.update_obstat(gen_obstat)
         self._push_obstat()
 
+    def noise_slice_flat(self, idx: int) -> np.ndarray:
+        """Noise slice at ``idx`` permuted into the FLAT state_dict layout.
+
+        On the engine path noise element t perturbs forward-layout parameter
+        t, so reconstructing a perturbed individual for ``Policy.pheno``
+        (flat layout) needs the inverse permutation.
+        """
+        fwd = self.nt.get(int(idx), self.n)
+        flat = torch.empty_like(fwd)
+        flat[self.perm] = fwd
+        return flat.cpu().numpy()
+
     def sync_host(self, light: bool = False):
         """Mirror device truth into the host Policy (checkpoint compatibility).
 
