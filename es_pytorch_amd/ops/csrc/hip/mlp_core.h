@@ -84,25 +84,42 @@ __device__ __forceinline__ float* mlp_layers(const uint16_t* __restrict__ wb,
       for (int q = 0; q < 8; ++q) acc[q] = 0.0f;
       if (ip < PART) {
         const uint16_t* wcol = Wt + (oi << 3);
-        // 4-deep manual unroll: keeps 4 HBM loads in flight per thread.
-        // (hipcc alone emits 1 load + s_waitcnt vmcnt(0) per iteration —
-        // measured ~5x slower, pure latency-bound.)
+        // Software-pipelined 4-wide register double-buffer: the NEXT block's
+        // 4 loads issue before the CURRENT block's FMAs, so ~4-8 HBM loads
+        // stay in flight across the loop back-edge. (hipcc alone emits one
+        // load + s_waitcnt vmcnt(0) per iteration — measured ~5x slower; a
+        // plain 4-unroll still drains vmcnt to 0 at every back-edge —
+        // SQ_WAIT_ANY measured at 71% of wave cycles.)
+        typedef uint32_t u32x4v __attribute__((ext_vector_type(4)));
+        auto ld = [&](int i) {
+          // nt: each member's weights are read by exactly one CU per step —
+          // keep them out of L2 so the shared dynamics matrix stays resident
+          u32x4v v = __builtin_nontemporal_load(
+              reinterpret_cast<const u32x4v*>(wcol + (int64_t)i * O));
+          uint4 w;
+          w.x = v.x; w.y = v.y; w.z = v.z; w.w = v.w;
+          return w;
+        };
         int i = ip;
         const int step4 = PART * 4;
-        for (; i + 3 * PART < I; i += step4) {
-          const uint4 w0 = *reinterpret_cast<const uint4*>(wcol + (int64_t)i * O);
-          const uint4 w1 = *reinterpret_cast<const uint4*>(wcol + (int64_t)(i + PART) * O);
-          const uint4 w2 = *reinterpret_cast<const uint4*>(wcol + (int64_t)(i + 2 * PART) * O);
-          const uint4 w3 = *reinterpret_cast<const uint4*>(wcol + (int64_t)(i + 3 * PART) * O);
-          bf8_fma(w0, x[i], acc);
-          bf8_fma(w1, x[i + PART], acc);
-          bf8_fma(w2, x[i + 2 * PART], acc);
-          bf8_fma(w3, x[i + 3 * PART], acc);
+        if (i + 3 * PART < I) {
+          uint4 c0 = ld(i), c1 = ld(i + PART), c2 = ld(i + 2 * PART), c3 = ld(i + 3 * PART);
+          for (; i + 7 * PART < I; i += step4) {
+            const uint4 n0 = ld(i + 4 * PART), n1 = ld(i + 5 * PART),
+                        n2 = ld(i + 6 * PART), n3 = ld(i + 7 * PART);
+            bf8_fma(c0, x[i], acc);
+            bf8_fma(c1, x[i + PART], acc);
+            bf8_fma(c2, x[i + 2 * PART], acc);
+            bf8_fma(c3, x[i + 3 * PART], acc);
+            c0 = n0; c1 = n1; c2 = n2; c3 = n3;
+          }
+          bf8_fma(c0, x[i], acc);
+          bf8_fma(c1, x[i + PART], acc);
+          bf8_fma(c2, x[i + 2 * PART], acc);
+          bf8_fma(c3, x[i + 3 * PART], acc);
+          i += step4;
         }
-        for (; i < I; i += PART) {
-          const uint4 w = *reinterpret_cast<const uint4*>(wcol + (int64_t)i * O);
-          bf8_fma(w, x[i], acc);
-        }
+        for (; i < I; i += PART) bf8_fma(ld(i), x[i], acc);
 #pragma unroll
         for (int q = 0; q < 8; ++q) partial[(ip * OCT + oi) * 8 + q] = acc[q];
       }

@@ -89,29 +89,36 @@ loco_step_kernel(const uint16_t* __restrict__ weights, const float* __restrict__
       float acc[4] = {0, 0, 0, 0};
       if (ip < PART) {
         const float* acol = Am + (oi << 2);
-        // 4-deep unroll: keep several L2 loads in flight (see mlp_core.h)
-        int i = ip;
-        const int step4 = PART * 4;
+        // software-pipelined register double-buffer (see mlp_core.h)
+        auto ld = [&](int i) {
+          return *reinterpret_cast<const float4*>(acol + (int64_t)i * S);
+        };
         auto f4 = [&](const float4 w, const float xi) {
           acc[0] = fmaf(w.x, xi, acc[0]);
           acc[1] = fmaf(w.y, xi, acc[1]);
           acc[2] = fmaf(w.z, xi, acc[2]);
           acc[3] = fmaf(w.w, xi, acc[3]);
         };
-        for (; i + 3 * PART < S; i += step4) {
-          const float4 w0 = *reinterpret_cast<const float4*>(acol + (int64_t)i * S);
-          const float4 w1 = *reinterpret_cast<const float4*>(acol + (int64_t)(i + PART) * S);
-          const float4 w2 = *reinterpret_cast<const float4*>(acol + (int64_t)(i + 2 * PART) * S);
-          const float4 w3 = *reinterpret_cast<const float4*>(acol + (int64_t)(i + 3 * PART) * S);
-          f4(w0, raws[i]);
-          f4(w1, raws[i + PART]);
-          f4(w2, raws[i + 2 * PART]);
-          f4(w3, raws[i + 3 * PART]);
+        int i = ip;
+        const int step4 = PART * 4;
+        if (i + 3 * PART < S) {
+          float4 c0 = ld(i), c1 = ld(i + PART), c2 = ld(i + 2 * PART), c3 = ld(i + 3 * PART);
+          for (; i + 7 * PART < S; i += step4) {
+            const float4 n0 = ld(i + 4 * PART), n1 = ld(i + 5 * PART),
+                         n2 = ld(i + 6 * PART), n3 = ld(i + 7 * PART);
+            f4(c0, raws[i]);
+            f4(c1, raws[i + PART]);
+            f4(c2, raws[i + 2 * PART]);
+            f4(c3, raws[i + 3 * PART]);
+            c0 = n0; c1 = n1; c2 = n2; c3 = n3;
+          }
+          f4(c0, raws[i]);
+          f4(c1, raws[i + PART]);
+          f4(c2, raws[i + 2 * PART]);
+          f4(c3, raws[i + 3 * PART]);
+          i += step4;
         }
-        for (; i < S; i += PART) {
-          const float4 w = *reinterpret_cast<const float4*>(acol + (int64_t)i * S);
-          f4(w, raws[i]);
-        }
+        for (; i < S; i += PART) f4(ld(i), raws[i]);
 #pragma unroll
         for (int q = 0; q < 4; ++q) partial[(ip * OCT + oi) * 4 + q] = acc[q];
       }

@@ -85,7 +85,7 @@ def test_mlp_fwd_kernel(dev):
     ops.check(ops.hip().es_mlp_fwd(actions.data_ptr(), obs.data_ptr(), weights.data_ptr(),
                                    obmean.data_ptr(), obstd.data_ptr(),
                                    dims_arr.ctypes.data, len(dims), None, 0, B,
-                                   5.0, 0.0, stride, 1, B, _stream(dev)), "mlp_fwd")
+                                   5.0, None, stride, 1, B, _stream(dev)), "mlp_fwd")
     torch.cuda.synchronize()
     ref = _torch_mlp_ref(obs, weights, dims, obmean, obstd, 5.0)
     assert torch.allclose(actions, ref, atol=2e-2, rtol=2e-2), \
@@ -109,7 +109,7 @@ def test_mlp_fwd_odd_output_dim(dev):
     ops.check(ops.hip().es_mlp_fwd(actions.data_ptr(), obs.data_ptr(), weights.data_ptr(),
                                    obmean.data_ptr(), obstd.data_ptr(),
                                    dims_arr.ctypes.data, len(dims), None, 0, B,
-                                   5.0, 0.0, stride, 1, B, _stream(dev)), "mlp_fwd")
+                                   5.0, None, stride, 1, B, _stream(dev)), "mlp_fwd")
     torch.cuda.synchronize()
     ref = _torch_mlp_ref(obs, weights, dims, obmean, obstd, 5.0)
     assert torch.allclose(actions, ref, atol=2e-2, rtol=2e-2)
@@ -130,11 +130,13 @@ def test_mlp_fwd_action_noise_statistics(dev):
     a2 = torch.empty(B, 2, device=dev)
     dims_arr = np.array(dims, dtype=np.int32)
     seed = torch.tensor([123], dtype=torch.int64, device=dev)
+    acstd = torch.tensor([0.5], dtype=torch.float32, device=dev)
     for out, salt in ((a1, 5), (a2, 5)):
         ops.check(ops.hip().es_mlp_fwd(out.data_ptr(), obs.data_ptr(), weights.data_ptr(),
                                        obmean.data_ptr(), obstd.data_ptr(),
                                        dims_arr.ctypes.data, len(dims), seed.data_ptr(),
-                                       salt, B, 5.0, 0.5, stride, 1, B, _stream(dev)), "mlp_fwd")
+                                       salt, B, 5.0, acstd.data_ptr(), stride, 1, B,
+                                       _stream(dev)), "mlp_fwd")
     torch.cuda.synchronize()
     assert torch.equal(a1, a2)  # same (seed, salt) -> same noise
     noise = a1.flatten()
@@ -144,7 +146,8 @@ def test_mlp_fwd_action_noise_statistics(dev):
     ops.check(ops.hip().es_mlp_fwd(a3.data_ptr(), obs.data_ptr(), weights.data_ptr(),
                                    obmean.data_ptr(), obstd.data_ptr(),
                                    dims_arr.ctypes.data, len(dims), seed.data_ptr(),
-                                   6, B, 5.0, 0.5, stride, 1, B, _stream(dev)), "mlp_fwd")
+                                   6, B, 5.0, acstd.data_ptr(), stride, 1, B,
+                                   _stream(dev)), "mlp_fwd")
     torch.cuda.synchronize()
     assert not torch.equal(a1, a3)  # different salt -> different noise
 
