@@ -417,6 +417,13 @@ class GpuEngine:
         flat = torch.empty(self.n, dtype=torch.float32, device=self.device)
         flat[self.perm] = self.theta
         self.policy.flat_params[:] = flat.cpu().numpy()
+        # failure detection: a diverged update (NaN/inf params) must stop the
+        # run loudly instead of silently training garbage for hours
+        if not np.isfinite(self.policy.flat_params).all():
+            raise FloatingPointError(
+                f"non-finite parameters after generation {self.gen} "
+                f"(lr={self.policy.optim.lr}, std={self.policy.std}) — "
+                "lower lr/noise std or check fitness scaling")
         if not light:
             opt = self.policy.optim
             inv = torch.empty_like(flat)

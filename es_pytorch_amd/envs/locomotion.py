@@ -78,15 +78,17 @@ class SyntheticLocomotion(BatchedEnv):
         self._steps = torch.zeros(batch, device=d)
 
     def reset(self, seed: Optional[int] = None) -> torch.Tensor:
-        g = torch.Generator(device="cpu")
+        # generate directly on device (avoids a host randn + H2D per generation)
+        g = torch.Generator(device=self.device)
         if seed is not None:
             g.manual_seed(int(seed))
         # in-place into persistent buffers: hipGraph-capture-safe
-        self.s.copy_((torch.randn(self.batch, self.sdim, generator=g) * 0.1).to(self.device))
+        torch.randn(self.batch, self.sdim, generator=g, device=self.device,
+                    out=self.s).mul_(0.1)
         self.pos.zero_()
         self._steps.zero_()
         if self.goal_conditioned:
-            ang = (torch.rand(self.batch, generator=g) * 2 * np.pi).to(self.device)
+            ang = torch.rand(self.batch, generator=g, device=self.device) * 2 * np.pi
             r = 15.0
             self.goal.copy_(torch.stack([r * torch.cos(ang), r * torch.sin(ang)], dim=1))
         return self._obs()

@@ -325,3 +325,53 @@ def test_fused_loco_termination(dev):
     # horizon, so rewards only match loosely; the bookkeeping (steps, freeze)
     # is what this test pins down
     np.testing.assert_allclose(rew_a[same], rew_b[same], rtol=5e-2, atol=5e-2)
+
+
+def test_engine_checkpoint_resume(dev, tmp_path):
+    """Engine -> reference-format pickle -> fresh engine: training state
+    (params, Adam moments, obstat) survives the round trip."""
+    import os
+
+    import numpy as np
+    from es_pytorch_amd.config import AttrDict
+    from es_pytorch_amd.core.engine import GpuEngine
+    from es_pytorch_amd.core.noisetable import NoiseTable
+    from es_pytorch_amd.core.policy import Policy
+    from es_pytorch_amd.envs import make_batched
+    from es_pytorch_amd.nn.nn import FeedForward
+    from es_pytorch_amd.nn.optimizers import Adam
+    from es_pytorch_amd.parallel.comm import Comm
+    from es_pytorch_amd.utils.rankers import CenteredRanker
+
+    torch.manual_seed(9)
+    comm = Comm(dev)
+    cfg = AttrDict({"env": {"name": "Hopper-v3", "max_steps": 20},
+                    "noise": {"tbl_size": 500_000, "std": 0.02},
+                    "policy": {"layer_sizes": [32], "ac_std": 0.0, "l2coeff": 0.005,
+                               "lr": 0.01, "ob_clip": 5, "save_obs_chance": 1.0},
+                    "general": {"policies_per_gen": 8, "batch_size": 500, "seed": 1}})
+    env = make_batched("Hopper-v3", 9, dev, max_steps=20, terminate_on_fall=False)
+    nn = FeedForward([32], torch.nn.Tanh(), env, 0.0, 5)
+    policy = Policy(nn, 0.02, Adam(len(Policy.get_flat(nn)), 0.01))
+    nt = NoiseTable.create_shared(comm, 500_000, len(policy), seed=2, device=dev)
+    rs = np.random.RandomState(7)
+    eng = GpuEngine(cfg, comm, policy, nt, env, rs, use_graph=False)
+    ranker = CenteredRanker()
+    for _ in range(2):
+        tr, obstat = eng.step(ranker)
+        eng.update_obstat(obstat)
+    eng.sync_host()  # full sync incl. optimizer moments
+    policy.save(str(tmp_path), "ckpt")
+
+    loaded = Policy.load(os.path.join(str(tmp_path), "policy-ckpt"))
+    np.testing.assert_array_equal(loaded.flat_params, policy.flat_params)
+    np.testing.assert_array_equal(loaded.optim.m, policy.optim.m)
+    assert loaded.optim.t == policy.optim.t == 2
+    assert loaded.obstat.count == policy.obstat.count
+
+    eng2 = GpuEngine(cfg, comm, loaded, nt, env, rs, use_graph=False)
+    # device state reconstructed from the checkpoint
+    np.testing.assert_allclose(eng2.theta.cpu().numpy(), eng.theta.cpu().numpy())
+    np.testing.assert_allclose(eng2.m.cpu().numpy(), eng.m.cpu().numpy(), atol=1e-7)
+    eng2.step(ranker)  # resumes without error
+    assert np.isfinite(loaded.flat_params).all()
