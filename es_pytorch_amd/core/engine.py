@@ -98,6 +98,8 @@ class GpuEngine:
         assert env.batch == self.B, f"env batch {env.batch} != engine batch {self.B}"
 
         self.dims = policy._module.layer_dims()
+        # K9: FFBinned policies emit adim*bins logits decoded in-kernel
+        self.bins = int(getattr(policy._module, "bins", 0))
         self.n = int(np.sum([I * O + O for I, O in zip(self.dims[:-1], self.dims[1:])]))
         assert self.n == len(policy), (self.n, len(policy))
         self.perm = forward_perm(self.dims).to(self.device)  # fwd idx -> flat idx
@@ -122,6 +124,14 @@ class GpuEngine:
         # device-resident mutable scalars read by kernels (graph-replay-safe)
         self.acstd_dev = torch.zeros(1, dtype=torch.float32, device=d)
 
+        if self.bins > 1:
+            self.alow_dev = torch.from_numpy(
+                np.asarray(env.action_space.low, dtype=np.float32)).to(d)
+            self.arange_dev = torch.from_numpy(np.asarray(
+                env.action_space.high - env.action_space.low, dtype=np.float32)).to(d)
+        else:
+            self.alow_dev = self.arange_dev = None
+
         D = env.ob_dim
         self.obmean = torch.zeros(D, dtype=torch.float32, device=d)
         self.obstd = torch.ones(D, dtype=torch.float32, device=d)
@@ -144,6 +154,10 @@ class GpuEngine:
         self.gen = 0
         self._graph: Optional[torch.cuda.CUDAGraph] = None
         self.timings = {}
+        # side stream: next generation's noise-offset upload overlaps the
+        # fitness all-gather (BASELINE north star)
+        self._side = torch.cuda.Stream(self.device) if self.device.type == "cuda" else None
+        self._offs_prefetched = False
 
         # fused single-kernel rollout step for the locomotion envs: policy
         # forward + dynamics + bookkeeping in one launch per step
@@ -172,7 +186,10 @@ class GpuEngine:
             self.obmean.data_ptr(), self.obstd.data_ptr(),
             self.dims_arr.ctypes.data, len(self.dims_arr), self.seed_dev.data_ptr(),
             salt, self.B, float(self.policy._module.ob_clip), self.acstd_dev.data_ptr(),
-            self.row_stride, 1, self.B - 1, self._stream()), "es_mlp_fwd")
+            self.row_stride, 1, self.B - 1, self.bins,
+            self.alow_dev.data_ptr() if self.alow_dev is not None else None,
+            self.arange_dev.data_ptr() if self.arange_dev is not None else None,
+            self._stream()), "es_mlp_fwd")
         return self.actions
 
     def _loco_step(self, t: int):
@@ -191,7 +208,7 @@ class GpuEngine:
             self.member_steps.data_ptr(), self.behv.data_ptr(),
             self.mo_sum.data_ptr(), self.mo_sumsq.data_ptr(),
             self.B, env.sdim, env.ac_dim, int(env.goal_conditioned),
-            int(env.terminate_on_fall), self.B - 1,
+            int(env.terminate_on_fall), self.B - 1, self.bins,
             float(env.leak), float(env.ctrl_cost), float(env.alive_bonus),
             float(env.fall_threshold), float(env.dt), self._stream()), "es_loco_step")
 
@@ -279,9 +296,12 @@ class GpuEngine:
         """Run one full generation; mirrors ``es.step`` semantics (``es.py:23-51``)."""
         t0 = time.perf_counter()
         cfg = self.cfg
-        offs = self.nt.sample_idxs(self.rs, self.pairs)
-        self.offsets[:self.pairs].copy_(torch.from_numpy(offs).to(self.device))
-        self.offsets[self.pairs:2 * self.pairs].copy_(self.offsets[:self.pairs])
+        if self._offs_prefetched:
+            # pheno (main stream) must see the side-stream offset upload
+            torch.cuda.current_stream(self.device).wait_stream(self._side)
+        else:
+            self._upload_offsets()
+        self._offs_prefetched = False
         self.seed_dev.fill_(int(self.rs.randint(0, 2 ** 31)))
         self.acstd_dev.fill_(float(getattr(self.policy._module, "_action_std", 0.0)))
         chance = float(cfg.policy.get("save_obs_chance", 1.0))
@@ -305,6 +325,13 @@ class GpuEngine:
         if self.device.type == "cuda":
             torch.cuda.synchronize(self.device)
         t2 = time.perf_counter()
+        # overlap: this generation's offsets are already captured in `rows`,
+        # so the NEXT generation's noise draw uploads on a side stream while
+        # the fitness all-gather is in flight over xGMI
+        if self._side is not None:
+            with torch.cuda.stream(self._side):
+                self._upload_offsets()
+            self._offs_prefetched = True
         all_rows = self.comm.allgather_rows(rows).cpu().numpy()
         # evaluated-episode steps only (the reference's es.py:79 counts the
         # pos+neg rollouts, not the noiseless eval)
@@ -355,6 +382,13 @@ class GpuEngine:
                         "update_s": t4 - t3, "gen_s": time.perf_counter() - t0,
                         "env_steps": steps}
         return noiseless, gen_obstat
+
+    def _upload_offsets(self):
+        """Sample this rank's antithetic noise offsets and upload to device."""
+        offs = self.nt.sample_idxs(self.rs, self.pairs)
+        self.offsets[:self.pairs].copy_(torch.from_numpy(offs).to(self.device,
+                                                                  non_blocking=True))
+        self.offsets[self.pairs:2 * self.pairs].copy_(self.offsets[:self.pairs])
 
     # ------------------------------------------------------------- update
     def _optim_step(self, n_ranked: float, l2: float):

@@ -14,7 +14,8 @@ mlp_fwd_kernel(float* __restrict__ actions, const float* __restrict__ obs,
                const float* __restrict__ obstd, MlpShape sh, float ob_clip,
                const float* __restrict__ ac_std_dev,
                const uint64_t* __restrict__ seed_dev, uint64_t salt, int64_t row_stride,
-               int act_final, int noiseless_from) {
+               int act_final, int noiseless_from, int bins,
+               const float* __restrict__ alow, const float* __restrict__ arange) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   float* bufA = reinterpret_cast<float*>(smem);
   float* bufB = bufA + sh.maxdim;
@@ -34,6 +35,22 @@ mlp_fwd_kernel(float* __restrict__ actions, const float* __restrict__ obs,
 
   const int A = sh.dims[sh.n_layers];
   const uint64_t seed = seed_dev ? (*seed_dev + salt) : salt;
+  if (bins > 1) {
+    // K9 binned-action decode (reference FFBinned, nn.py:111-117): per-dim
+    // argmax over bins, rescaled into [alow, alow+range]
+    const int adim = A / bins;
+    for (int d = tid; d < adim; d += blockDim.x) {
+      int best = 0;
+      float bv = out[d * bins];
+      for (int j = 1; j < bins; ++j) {
+        const float v = out[d * bins + j];
+        if (v > bv) { bv = v; best = j; }
+      }
+      actions[(int64_t)b * adim + d] =
+          alow[d] + arange[d] * (float)best / (float)(bins - 1);
+    }
+    return;
+  }
   // ac_std is read from device memory so decay schedules keep working under
   // hipGraph replay (kernel args are frozen at capture time)
   const float ac_std = ac_std_dev ? *ac_std_dev : 0.0f;
@@ -51,7 +68,8 @@ extern "C" int es_mlp_fwd(void* actions, const void* obs, const void* weights,
                           const void* obmean, const void* obstd, const int32_t* dims_host,
                           int32_t ndims, const void* seed_dev, uint64_t salt, int32_t n_pop,
                           float ob_clip, const void* ac_std_dev, int64_t row_stride,
-                          int32_t act_final, int32_t noiseless_from, void* stream) {
+                          int32_t act_final, int32_t noiseless_from, int32_t bins,
+                          const void* alow, const void* arange, void* stream) {
   MlpShape sh;
   int rc = mlp_shape_init(&sh, dims_host, ndims, row_stride);
   if (rc) return rc;
@@ -59,7 +77,8 @@ extern "C" int es_mlp_fwd(void* actions, const void* obs, const void* weights,
                    (hipStream_t)stream>>>(
       (float*)actions, (const float*)obs, (const uint16_t*)weights, (const float*)obmean,
       (const float*)obstd, sh, ob_clip, (const float*)ac_std_dev,
-      (const uint64_t*)seed_dev, salt, row_stride, act_final, noiseless_from);
+      (const uint64_t*)seed_dev, salt, row_stride, act_final, noiseless_from, bins,
+      (const float*)alow, (const float*)arange);
   ES_CHECK_LAUNCH();
   return 0;
 }

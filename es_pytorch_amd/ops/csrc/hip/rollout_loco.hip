@@ -22,6 +22,7 @@ struct LocoArgs {
   int goal;         // goal-conditioned flag
   int terminate;    // terminate_on_fall
   int noiseless_from;  // members >= this index get no action noise
+  int bins;            // >1: K9 binned-action decode (FFBinned)
   float leak, ctrl, alive_bonus, fall_thr, dt, ob_clip;
   uint64_t salt;
   int64_t row_stride;
@@ -71,7 +72,17 @@ loco_step_kernel(const uint16_t* __restrict__ weights, const float* __restrict__
   const float* aout = mlp_layers(wb, sh, bufA, bufB, partial, tid, nth, 1);
   const uint64_t seed = seed_dev ? (*seed_dev + la.salt) : la.salt;
   const float ac_std = ac_std_dev ? *ac_std_dev : 0.0f;  // device-read: graph-safe decay
-  if (tid < A) {
+  if (la.bins > 1) {
+    // K9 binned decode (FFBinned): per-dim argmax over bins -> [-1, 1]
+    if (tid < A) {
+      const float* row = aout + tid * la.bins;
+      int best = 0;
+      float bv = row[0];
+      for (int j = 1; j < la.bins; ++j)
+        if (row[j] > bv) { bv = row[j]; best = j; }
+      abuf[tid] = -1.0f + 2.0f * (float)best / (float)(la.bins - 1);
+    }
+  } else if (tid < A) {
     float a = aout[tid];
     if (ac_std != 0.0f && b < la.noiseless_from)
       a += ac_std * es_actnoise(seed, (uint64_t)b * A + tid);
@@ -232,18 +243,19 @@ extern "C" int es_loco_step(const void* weights, const void* obmean, const void*
                             const void* wy, const void* wh, void* alive, void* rew_total,
                             void* member_steps, void* behv, void* mo_sum, void* mo_sumsq,
                             int32_t n_pop, int32_t sdim, int32_t adim, int32_t goal_flag,
-                            int32_t terminate, int32_t noiseless_from, float leak,
-                            float ctrl, float alive_bonus, float fall_thr, float dt,
-                            void* stream) {
+                            int32_t terminate, int32_t noiseless_from, int32_t bins,
+                            float leak, float ctrl, float alive_bonus, float fall_thr,
+                            float dt, void* stream) {
   MlpShape sh;
   int rc = mlp_shape_init(&sh, dims_host, ndims, row_stride);
   if (rc) return rc;
   if (adim > 64 || sdim > ES_MAXDIM) return -103;
   if (sh.dims[0] != sdim + (goal_flag ? 2 : 0)) return -104;
-  if (sh.dims[sh.n_layers] != adim) return -105;
+  const int out_dim = bins > 1 ? adim * bins : adim;
+  if (sh.dims[sh.n_layers] != out_dim) return -105;
   LocoArgs la;
   la.S = sdim; la.A = adim; la.D = sh.dims[0]; la.goal = goal_flag;
-  la.terminate = terminate; la.noiseless_from = noiseless_from;
+  la.terminate = terminate; la.noiseless_from = noiseless_from; la.bins = bins;
   la.leak = leak; la.ctrl = ctrl; la.alive_bonus = alive_bonus; la.fall_thr = fall_thr;
   la.dt = dt; la.ob_clip = ob_clip; la.salt = salt;
   la.row_stride = row_stride;

@@ -85,7 +85,7 @@ def test_mlp_fwd_kernel(dev):
     ops.check(ops.hip().es_mlp_fwd(actions.data_ptr(), obs.data_ptr(), weights.data_ptr(),
                                    obmean.data_ptr(), obstd.data_ptr(),
                                    dims_arr.ctypes.data, len(dims), None, 0, B,
-                                   5.0, None, stride, 1, B, _stream(dev)), "mlp_fwd")
+                                   5.0, None, stride, 1, B, 0, None, None, _stream(dev)), "mlp_fwd")
     torch.cuda.synchronize()
     ref = _torch_mlp_ref(obs, weights, dims, obmean, obstd, 5.0)
     assert torch.allclose(actions, ref, atol=2e-2, rtol=2e-2), \
@@ -109,7 +109,7 @@ def test_mlp_fwd_odd_output_dim(dev):
     ops.check(ops.hip().es_mlp_fwd(actions.data_ptr(), obs.data_ptr(), weights.data_ptr(),
                                    obmean.data_ptr(), obstd.data_ptr(),
                                    dims_arr.ctypes.data, len(dims), None, 0, B,
-                                   5.0, None, stride, 1, B, _stream(dev)), "mlp_fwd")
+                                   5.0, None, stride, 1, B, 0, None, None, _stream(dev)), "mlp_fwd")
     torch.cuda.synchronize()
     ref = _torch_mlp_ref(obs, weights, dims, obmean, obstd, 5.0)
     assert torch.allclose(actions, ref, atol=2e-2, rtol=2e-2)
@@ -136,7 +136,7 @@ def test_mlp_fwd_action_noise_statistics(dev):
                                        obmean.data_ptr(), obstd.data_ptr(),
                                        dims_arr.ctypes.data, len(dims), seed.data_ptr(),
                                        salt, B, 5.0, acstd.data_ptr(), stride, 1, B,
-                                       _stream(dev)), "mlp_fwd")
+                                       0, None, None, _stream(dev)), "mlp_fwd")
     torch.cuda.synchronize()
     assert torch.equal(a1, a2)  # same (seed, salt) -> same noise
     noise = a1.flatten()
@@ -147,7 +147,7 @@ def test_mlp_fwd_action_noise_statistics(dev):
                                    obmean.data_ptr(), obstd.data_ptr(),
                                    dims_arr.ctypes.data, len(dims), seed.data_ptr(),
                                    6, B, 5.0, acstd.data_ptr(), stride, 1, B,
-                                   _stream(dev)), "mlp_fwd")
+                                   0, None, None, _stream(dev)), "mlp_fwd")
     torch.cuda.synchronize()
     assert not torch.equal(a1, a3)  # different salt -> different noise
 
@@ -375,3 +375,39 @@ def test_engine_checkpoint_resume(dev, tmp_path):
     np.testing.assert_allclose(eng2.m.cpu().numpy(), eng.m.cpu().numpy(), atol=1e-7)
     eng2.step(ranker)  # resumes without error
     assert np.isfinite(loaded.flat_params).all()
+
+
+def test_binned_policy_engine(dev):
+    """K9 binned-action decode: FFBinned on the engine, fused vs torch-path
+    parity and agreement with the episodic FFBinned forward."""
+    import numpy as np
+    from es_pytorch_amd.config import AttrDict
+    from es_pytorch_amd.core.engine import GpuEngine
+    from es_pytorch_amd.core.noisetable import NoiseTable
+    from es_pytorch_amd.core.policy import Policy
+    from es_pytorch_amd.envs import make_batched
+    from es_pytorch_amd.nn.nn import FFBinned
+    from es_pytorch_amd.nn.optimizers import Adam
+    from es_pytorch_amd.parallel.comm import Comm
+    from es_pytorch_amd.utils.rankers import CenteredRanker
+
+    fits = {}
+    for fused in (False, True):
+        torch.manual_seed(6)
+        comm = Comm(dev)
+        cfg = AttrDict({"env": {"name": "Hopper-v3", "max_steps": 25},
+                        "noise": {"tbl_size": 500_000, "std": 0.05},
+                        "policy": {"layer_sizes": [32], "ac_std": 0.0, "l2coeff": 0.005,
+                                   "lr": 0.01, "ob_clip": 5, "save_obs_chance": 1.0},
+                        "general": {"policies_per_gen": 8, "batch_size": 500, "seed": 1}})
+        env = make_batched("Hopper-v3", 9, dev, max_steps=25, terminate_on_fall=False)
+        nn = FFBinned([32], torch.nn.Tanh(), env, n_bins=5, ob_clip=5)
+        policy = Policy(nn, 0.05, Adam(len(Policy.get_flat(nn)), 0.01))
+        nt = NoiseTable.create_shared(comm, 500_000, len(policy), seed=4, device=dev)
+        rs = np.random.RandomState(13)
+        eng = GpuEngine(cfg, comm, policy, nt, env, rs, use_graph=False, fused=fused)
+        assert eng.bins == 5
+        ranker = CenteredRanker()
+        eng.step(ranker)
+        fits[fused] = np.concatenate([ranker.fits_pos, ranker.fits_neg]).ravel()
+    np.testing.assert_allclose(fits[False], fits[True], rtol=1e-3, atol=1e-2)
