@@ -193,7 +193,9 @@ class GpuEngine:
         return self.actions
 
     def _loco_step(self, t: int):
-        """One fused rollout step (rollout_loco.hip): forward + env + bookkeeping."""
+        """One fused rollout step for the 2*pairs perturbed members; the
+        noiseless member runs as a side-stream whole-episode kernel so the
+        main grid stays an exact multiple of the CU slot count."""
         env = self.env
         goal_ptr = env.goal.data_ptr() if env.goal_conditioned else None
         ops.check(ops.hip().es_loco_step(
@@ -207,10 +209,31 @@ class GpuEngine:
             self.alive.data_ptr(), self.rew_total.data_ptr(),
             self.member_steps.data_ptr(), self.behv.data_ptr(),
             self.mo_sum.data_ptr(), self.mo_sumsq.data_ptr(),
-            self.B, env.sdim, env.ac_dim, int(env.goal_conditioned),
+            self.B - 1, env.sdim, env.ac_dim, int(env.goal_conditioned),
             int(env.terminate_on_fall), self.B - 1, self.bins,
             float(env.leak), float(env.ctrl_cost), float(env.alive_bonus),
             float(env.fall_threshold), float(env.dt), self._stream()), "es_loco_step")
+
+    def _loco_noiseless_episode(self):
+        """Whole noiseless episode (member B-1) in one kernel launch."""
+        env = self.env
+        goal_ptr = env.goal.data_ptr() if env.goal_conditioned else None
+        ops.check(ops.hip().es_loco_episode(
+            self.weights.data_ptr(), self.obmean.data_ptr(), self.obstd.data_ptr(),
+            self.dims_arr.ctypes.data, len(self.dims_arr), self.seed_dev.data_ptr(),
+            self.max_steps, float(self.policy._module.ob_clip),
+            self.acstd_dev.data_ptr(), self.row_stride,
+            env.s.data_ptr(), env.pos.data_ptr(), goal_ptr,
+            env.A.data_ptr(), env.B.data_ptr(), env.b0.data_ptr(),
+            env.wv.data_ptr(), env.wa.data_ptr(), env.wy.data_ptr(), env.wh.data_ptr(),
+            self.alive.data_ptr(), self.rew_total.data_ptr(),
+            self.member_steps.data_ptr(), self.behv.data_ptr(),
+            self.mo_sum.data_ptr(), self.mo_sumsq.data_ptr(),
+            self.B - 1, 1, env.sdim, env.ac_dim, int(env.goal_conditioned),
+            int(env.terminate_on_fall), 0, self.bins,
+            float(env.leak), float(env.ctrl_cost), float(env.alive_bonus),
+            float(env.fall_threshold), float(env.dt), self._stream()),
+            "es_loco_episode")
 
     # ------------------------------------------------------------- rollout
     def _step_body(self, t: int):
@@ -231,6 +254,21 @@ class GpuEngine:
         self.alive.mul_(1.0 - done.float())
         self.obs_buf.copy_(ob)
 
+    def _rollout_body(self):
+        if self.fused:
+            # noiseless member: one whole-episode kernel on a side stream,
+            # concurrent with the per-step population kernels
+            main = torch.cuda.current_stream(self.device)
+            self._side.wait_stream(main)
+            with torch.cuda.stream(self._side):
+                self._loco_noiseless_episode()
+            for t in range(self.max_steps):
+                self._loco_step(t)
+            main.wait_stream(self._side)
+        else:
+            for t in range(self.max_steps):
+                self._step_body(t)
+
     def _rollout(self):
         body = self._loco_step if self.fused else self._step_body
         if self.use_graph:
@@ -241,18 +279,18 @@ class GpuEngine:
                 with torch.cuda.stream(s):
                     for t in range(3):
                         body(t)
+                    if self.fused:
+                        self._loco_noiseless_episode()  # warm the episode kernel
                 torch.cuda.current_stream(self.device).wait_stream(s)
                 # warmup dirtied the rollout state: restore it before capture
                 self._reset_rollout_state()
                 self.obs_buf.copy_(self.env.reset(self._gen_seed()))
                 self._graph = torch.cuda.CUDAGraph()
                 with torch.cuda.graph(self._graph):
-                    for t in range(self.max_steps):
-                        body(t)
+                    self._rollout_body()
             self._graph.replay()
             return
-        for t in range(self.max_steps):
-            body(t)
+        self._rollout_body()
 
     def _reset_rollout_state(self):
         self.alive.fill_(1.0)

@@ -103,6 +103,8 @@ __device__ __forceinline__ float* mlp_layers(const uint16_t* __restrict__ wb,
         int i = ip;
         const int step4 = PART * 4;
         if (i + 3 * PART < I) {
+          // depth-4 register double-buffer (depth-8 raised VGPRs to 111 and
+          // cost a resident block per CU — measured net negative)
           uint4 c0 = ld(i), c1 = ld(i + PART), c2 = ld(i + 2 * PART), c3 = ld(i + 3 * PART);
           for (; i + 7 * PART < I; i += step4) {
             const uint4 n0 = ld(i + 4 * PART), n1 = ld(i + 5 * PART),

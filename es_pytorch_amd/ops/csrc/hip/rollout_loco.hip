@@ -1,77 +1,71 @@
-// Fused rollout step for the synthetic locomotion envs: ONE kernel per env
-// step does, per population member (one workgroup each):
+// Fused rollout kernels for the synthetic locomotion envs.
 //
+// Per population member (one workgroup), one env step comprises:
 //   policy MLP forward (mlp_core.h scheme, bf16 weight streaming)
-//   + gaussian action noise (skipped for the noiseless slot, reference
-//     es.py:48 evaluates noiselessly with rs=None)
+//   + gaussian action noise or K9 binned decode (skipped / noise-free for
+//     the noiseless evaluation, reference es.py:48 passes rs=None)
 //   + env dynamics  s' = (1-leak) s + leak tanh(s A + a B + b0)
 //   + reward / positions / fall-termination (envs/locomotion.py semantics)
 //   + alive-masked bookkeeping: total reward, per-member steps, behaviour
 //     freeze, per-member observation sums for ObStat
 //
-// replacing ~25 small torch kernels + 1 forward launch per step (measured
-// ~470 us/step) with a single HBM-bandwidth-bound launch. The A matrix is
-// shared by all members (L2-resident); activations and state never leave
-// LDS; actions never touch HBM.
+// Two launch shapes share the body:
+//  * es_loco_step    — ONE step for a whole population (grid = pop). The
+//    engine replays max_steps of these from a hipGraph. Replaced ~25 small
+//    torch kernels + a forward launch per step (measured ~470 us/step).
+//  * es_loco_episode — a WHOLE episode for a few members (internal step
+//    loop). Used for the noiseless evaluation on a side stream so the main
+//    per-step grid stays an exact multiple of the CU slot count (a +1
+//    straggler block measured ~15-20% tail on every step).
 #include "mlp_core.h"
 
 struct LocoArgs {
-  int S;            // latent state dim
-  int A;            // action dim
-  int D;            // obs dim (= S, or S+2 goal-conditioned)
-  int goal;         // goal-conditioned flag
-  int terminate;    // terminate_on_fall
+  int S;               // latent state dim
+  int A;               // action dim
+  int D;               // obs dim (= S, or S+2 goal-conditioned)
+  int goal;            // goal-conditioned flag
+  int terminate;       // terminate_on_fall
   int noiseless_from;  // members >= this index get no action noise
   int bins;            // >1: K9 binned-action decode (FFBinned)
   float leak, ctrl, alive_bonus, fall_thr, dt, ob_clip;
-  uint64_t salt;
   int64_t row_stride;
 };
 
-__global__ void __launch_bounds__(256)
-loco_step_kernel(const uint16_t* __restrict__ weights, const float* __restrict__ obmean,
-                 const float* __restrict__ obstd, MlpShape sh, LocoArgs la,
-                 const float* __restrict__ ac_std_dev,
-                 const uint64_t* __restrict__ seed_dev,
-                 float* __restrict__ s_glob, float* __restrict__ pos,
-                 const float* __restrict__ goal, const float* __restrict__ Am,
-                 const float* __restrict__ Bm, const float* __restrict__ b0,
-                 const float* __restrict__ wv, const float* __restrict__ wa,
-                 const float* __restrict__ wy, const float* __restrict__ wh,
-                 float* __restrict__ alive, float* __restrict__ rew_total,
-                 float* __restrict__ member_steps, float* __restrict__ behv,
-                 float* __restrict__ mo_sum, float* __restrict__ mo_sumsq) {
-  extern __shared__ __attribute__((aligned(16))) char smem[];
-  float* bufA = reinterpret_cast<float*>(smem);
-  float* bufB = bufA + sh.maxdim;
-  float* partial = bufB + sh.maxdim;
-  float* raws = partial + 256 * 8;
-  float* abuf = raws + ((la.S + 3) & ~3);
-  float* sc = abuf + 64;  // [4]: alive_old broadcast
-  const int b = blockIdx.x;
+struct LocoPtrs {
+  const uint16_t* weights;
+  const float *obmean, *obstd, *ac_std_dev;
+  const uint64_t* seed_dev;
+  float *s_glob, *pos;
+  const float *goal, *Am, *Bm, *b0, *wv, *wa, *wy, *wh;
+  float *alive, *rew_total, *member_steps, *behv, *mo_sum, *mo_sumsq;
+};
+
+__device__ __forceinline__ void loco_step_body(
+    const MlpShape& sh, const LocoArgs& la, const LocoPtrs& P, int b, uint64_t salt,
+    float* bufA, float* bufB, float* partial, float* raws, float* abuf, float* sc) {
   const int tid = threadIdx.x;
   const int nth = blockDim.x;
   const int S = la.S, A = la.A;
-  float* sb = s_glob + (int64_t)b * S;
+  float* sb = P.s_glob + (int64_t)b * S;
 
-  // ---- build normalized obs in buf[0]; keep raw state in LDS -------------
+  // ---- build normalized obs; keep raw state in LDS -----------------------
   for (int i = tid; i < S; i += nth) {
     const float v = sb[i];
     raws[i] = v;
-    bufA[i] = fclampf((v - obmean[i]) / obstd[i], -la.ob_clip, la.ob_clip);
+    bufA[i] = fclampf((v - P.obmean[i]) / P.obstd[i], -la.ob_clip, la.ob_clip);
   }
   if (la.goal && tid < 2) {
-    const float rel = (goal[(int64_t)b * 2 + tid] - pos[(int64_t)b * 3 + tid]) * 0.1f;
-    bufA[S + tid] = fclampf((rel - obmean[S + tid]) / obstd[S + tid], -la.ob_clip,
+    const float rel = (P.goal[(int64_t)b * 2 + tid] - P.pos[(int64_t)b * 3 + tid]) * 0.1f;
+    bufA[S + tid] = fclampf((rel - P.obmean[S + tid]) / P.obstd[S + tid], -la.ob_clip,
                             la.ob_clip);
   }
   __syncthreads();
 
   // ---- policy forward ----------------------------------------------------
-  const uint16_t* wb = weights + (int64_t)b * la.row_stride;
+  const uint16_t* wb = P.weights + (int64_t)b * la.row_stride;
   const float* aout = mlp_layers(wb, sh, bufA, bufB, partial, tid, nth, 1);
-  const uint64_t seed = seed_dev ? (*seed_dev + la.salt) : la.salt;
-  const float ac_std = ac_std_dev ? *ac_std_dev : 0.0f;  // device-read: graph-safe decay
+  const uint64_t seed = P.seed_dev ? (*P.seed_dev + salt) : salt;
+  const float ac_std = P.ac_std_dev ? *P.ac_std_dev : 0.0f;
   if (la.bins > 1) {
     // K9 binned decode (FFBinned): per-dim argmax over bins -> [-1, 1]
     if (tid < A) {
@@ -99,7 +93,7 @@ loco_step_kernel(const uint16_t* __restrict__ weights, const float* __restrict__
       const int oi = tid % OCT, ip = tid / OCT;
       float acc[4] = {0, 0, 0, 0};
       if (ip < PART) {
-        const float* acol = Am + (oi << 2);
+        const float* acol = P.Am + (oi << 2);
         // software-pipelined register double-buffer (see mlp_core.h)
         auto ld = [&](int i) {
           return *reinterpret_cast<const float4*>(acol + (int64_t)i * S);
@@ -135,19 +129,19 @@ loco_step_kernel(const uint16_t* __restrict__ weights, const float* __restrict__
       }
       __syncthreads();
       for (int o = tid; o < S; o += nth) {
-        float p = b0[o];
+        float p = P.b0[o];
         const int oo = o >> 2, j = o & 3;
         for (int pp = 0; pp < PART; ++pp) p += partial[(pp * OCT + oo) * 4 + j];
-        for (int k = 0; k < A; ++k) p = fmaf(abuf[k], Bm[(int64_t)k * S + o], p);
+        for (int k = 0; k < A; ++k) p = fmaf(abuf[k], P.Bm[(int64_t)k * S + o], p);
         const float sn = (1.0f - la.leak) * raws[o] + la.leak * tanhf(p);
         bufA[o] = sn;
         sb[o] = sn;
       }
     } else {
       for (int o = tid; o < S; o += nth) {
-        float p = b0[o];
-        for (int i = 0; i < S; ++i) p = fmaf(raws[i], Am[(int64_t)i * S + o], p);
-        for (int k = 0; k < A; ++k) p = fmaf(abuf[k], Bm[(int64_t)k * S + o], p);
+        float p = P.b0[o];
+        for (int i = 0; i < S; ++i) p = fmaf(raws[i], P.Am[(int64_t)i * S + o], p);
+        for (int k = 0; k < A; ++k) p = fmaf(abuf[k], P.Bm[(int64_t)k * S + o], p);
         const float sn = (1.0f - la.leak) * raws[o] + la.leak * tanhf(p);
         bufA[o] = sn;
         sb[o] = sn;
@@ -161,12 +155,12 @@ loco_step_kernel(const uint16_t* __restrict__ weights, const float* __restrict__
     float p0 = 0, p1 = 0, p2 = 0, p3 = 0;
     for (int i = tid; i < S; i += nth) {
       const float sn = bufA[i];
-      p0 = fmaf(sn, wv[i], p0);
-      p1 = fmaf(sn, wy[i], p1);
-      p2 = fmaf(sn, wh[i], p2);
+      p0 = fmaf(sn, P.wv[i], p0);
+      p1 = fmaf(sn, P.wy[i], p1);
+      p2 = fmaf(sn, P.wh[i], p2);
     }
     for (int j = tid; j < A; j += nth) {
-      p0 = fmaf(0.5f * abuf[j], wa[j], p0);
+      p0 = fmaf(0.5f * abuf[j], P.wa[j], p0);
       p3 = fmaf(abuf[j], abuf[j], p3);
     }
     partial[tid] = p0;
@@ -188,30 +182,30 @@ loco_step_kernel(const uint16_t* __restrict__ weights, const float* __restrict__
   // ---- scalar bookkeeping (thread 0) -------------------------------------
   if (tid == 0) {
     const float vfwd = partial[0], vy = partial[256], h = partial[512], asq = partial[768];
-    const float alive_old = alive[b];
+    const float alive_old = P.alive[b];
     float rew;
     if (la.goal) {
-      const float rx = goal[(int64_t)b * 2 + 0] - pos[(int64_t)b * 3 + 0];
-      const float ry = goal[(int64_t)b * 2 + 1] - pos[(int64_t)b * 3 + 1];
+      const float rx = P.goal[(int64_t)b * 2 + 0] - P.pos[(int64_t)b * 3 + 0];
+      const float ry = P.goal[(int64_t)b * 2 + 1] - P.pos[(int64_t)b * 3 + 1];
       const float inv = 1.0f / (sqrtf(rx * rx + ry * ry) + 1e-6f);
       rew = vfwd * rx * inv + vy * ry * inv - la.ctrl * asq + la.alive_bonus;
     } else {
       rew = vfwd - la.ctrl * asq + la.alive_bonus;
     }
-    const float px = pos[(int64_t)b * 3 + 0] + la.dt * vfwd;
-    const float py = pos[(int64_t)b * 3 + 1] + la.dt * vy;
-    pos[(int64_t)b * 3 + 0] = px;
-    pos[(int64_t)b * 3 + 1] = py;
-    pos[(int64_t)b * 3 + 2] = h;
+    const float px = P.pos[(int64_t)b * 3 + 0] + la.dt * vfwd;
+    const float py = P.pos[(int64_t)b * 3 + 1] + la.dt * vy;
+    P.pos[(int64_t)b * 3 + 0] = px;
+    P.pos[(int64_t)b * 3 + 1] = py;
+    P.pos[(int64_t)b * 3 + 2] = h;
     const float done = (la.terminate && h < la.fall_thr) ? 1.0f : 0.0f;
-    rew_total[b] += rew * alive_old;
-    member_steps[b] += alive_old;
+    P.rew_total[b] += rew * alive_old;
+    P.member_steps[b] += alive_old;
     if (alive_old > 0.0f) {
-      behv[(int64_t)b * 3 + 0] = px;
-      behv[(int64_t)b * 3 + 1] = py;
-      behv[(int64_t)b * 3 + 2] = h;
+      P.behv[(int64_t)b * 3 + 0] = px;
+      P.behv[(int64_t)b * 3 + 1] = py;
+      P.behv[(int64_t)b * 3 + 2] = h;
     }
-    alive[b] = alive_old * (1.0f - done);
+    P.alive[b] = alive_old * (1.0f - done);
     sc[4] = alive_old;
   }
   __syncthreads();
@@ -219,19 +213,81 @@ loco_step_kernel(const uint16_t* __restrict__ weights, const float* __restrict__
   // ---- per-member obs statistics (post-step obs, alive-weighted) ---------
   const float w = sc[4];
   if (w > 0.0f) {
-    float* ms = mo_sum + (int64_t)b * la.D;
-    float* mq = mo_sumsq + (int64_t)b * la.D;
+    float* ms = P.mo_sum + (int64_t)b * la.D;
+    float* mq = P.mo_sumsq + (int64_t)b * la.D;
     for (int i = tid; i < S; i += nth) {
       const float o = bufA[i];
       ms[i] += o;
       mq[i] += o * o;
     }
     if (la.goal && tid < 2) {
-      const float rel = (goal[(int64_t)b * 2 + tid] - pos[(int64_t)b * 3 + tid]) * 0.1f;
+      const float rel = (P.goal[(int64_t)b * 2 + tid] - P.pos[(int64_t)b * 3 + tid]) * 0.1f;
       ms[S + tid] += rel;
       mq[S + tid] += rel * rel;
     }
   }
+  __syncthreads();  // LDS reuse safety for the episode kernel's next step
+}
+
+#define ES_LOCO_CARVE()                                          \
+  extern __shared__ __attribute__((aligned(16))) char smem[];    \
+  float* bufA = reinterpret_cast<float*>(smem);                  \
+  float* bufB = bufA + sh.maxdim;                                \
+  float* partial = bufB + sh.maxdim;                             \
+  float* raws = partial + 256 * 8;                               \
+  float* abuf = raws + ((la.S + 3) & ~3);                        \
+  float* sc = abuf + 64;
+
+__global__ void __launch_bounds__(256)
+loco_step_kernel(MlpShape sh, LocoArgs la, LocoPtrs P, uint64_t salt) {
+  ES_LOCO_CARVE();
+  loco_step_body(sh, la, P, blockIdx.x, salt, bufA, bufB, partial, raws, abuf, sc);
+}
+
+__global__ void __launch_bounds__(256)
+loco_episode_kernel(MlpShape sh, LocoArgs la, LocoPtrs P, int member_base, int n_steps) {
+  ES_LOCO_CARVE();
+  const int b = member_base + blockIdx.x;
+  for (int t = 1; t <= n_steps; ++t)
+    loco_step_body(sh, la, P, b, (uint64_t)t, bufA, bufB, partial, raws, abuf, sc);
+}
+
+static int loco_prepare(MlpShape* sh, LocoArgs* la, const int32_t* dims_host, int32_t ndims,
+                        int64_t row_stride, float ob_clip, int32_t sdim, int32_t adim,
+                        int32_t goal_flag, int32_t terminate, int32_t noiseless_from,
+                        int32_t bins, float leak, float ctrl, float alive_bonus,
+                        float fall_thr, float dt, unsigned* lds) {
+  int rc = mlp_shape_init(sh, dims_host, ndims, row_stride);
+  if (rc) return rc;
+  if (adim > 64 || sdim > ES_MAXDIM) return -103;
+  if (sh->dims[0] != sdim + (goal_flag ? 2 : 0)) return -104;
+  const int out_dim = bins > 1 ? adim * bins : adim;
+  if (sh->dims[sh->n_layers] != out_dim) return -105;
+  la->S = sdim; la->A = adim; la->D = sh->dims[0]; la->goal = goal_flag;
+  la->terminate = terminate; la->noiseless_from = noiseless_from; la->bins = bins;
+  la->leak = leak; la->ctrl = ctrl; la->alive_bonus = alive_bonus; la->fall_thr = fall_thr;
+  la->dt = dt; la->ob_clip = ob_clip; la->row_stride = row_stride;
+  *lds = (unsigned)(mlp_lds_bytes(sh->maxdim) + (((sdim + 3) & ~3) + 64 + 8) * 4);
+  return 0;
+}
+
+static LocoPtrs loco_ptrs(const void* weights, const void* obmean, const void* obstd,
+                          const void* ac_std_dev, const void* seed_dev, void* s_glob,
+                          void* pos, const void* goal, const void* Am, const void* Bm,
+                          const void* b0, const void* wv, const void* wa, const void* wy,
+                          const void* wh, void* alive, void* rew_total, void* member_steps,
+                          void* behv, void* mo_sum, void* mo_sumsq) {
+  LocoPtrs P;
+  P.weights = (const uint16_t*)weights;
+  P.obmean = (const float*)obmean; P.obstd = (const float*)obstd;
+  P.ac_std_dev = (const float*)ac_std_dev; P.seed_dev = (const uint64_t*)seed_dev;
+  P.s_glob = (float*)s_glob; P.pos = (float*)pos; P.goal = (const float*)goal;
+  P.Am = (const float*)Am; P.Bm = (const float*)Bm; P.b0 = (const float*)b0;
+  P.wv = (const float*)wv; P.wa = (const float*)wa; P.wy = (const float*)wy;
+  P.wh = (const float*)wh; P.alive = (float*)alive; P.rew_total = (float*)rew_total;
+  P.member_steps = (float*)member_steps; P.behv = (float*)behv;
+  P.mo_sum = (float*)mo_sum; P.mo_sumsq = (float*)mo_sumsq;
+  return P;
 }
 
 extern "C" int es_loco_step(const void* weights, const void* obmean, const void* obstd,
@@ -247,27 +303,46 @@ extern "C" int es_loco_step(const void* weights, const void* obmean, const void*
                             float leak, float ctrl, float alive_bonus, float fall_thr,
                             float dt, void* stream) {
   MlpShape sh;
-  int rc = mlp_shape_init(&sh, dims_host, ndims, row_stride);
-  if (rc) return rc;
-  if (adim > 64 || sdim > ES_MAXDIM) return -103;
-  if (sh.dims[0] != sdim + (goal_flag ? 2 : 0)) return -104;
-  const int out_dim = bins > 1 ? adim * bins : adim;
-  if (sh.dims[sh.n_layers] != out_dim) return -105;
   LocoArgs la;
-  la.S = sdim; la.A = adim; la.D = sh.dims[0]; la.goal = goal_flag;
-  la.terminate = terminate; la.noiseless_from = noiseless_from; la.bins = bins;
-  la.leak = leak; la.ctrl = ctrl; la.alive_bonus = alive_bonus; la.fall_thr = fall_thr;
-  la.dt = dt; la.ob_clip = ob_clip; la.salt = salt;
-  la.row_stride = row_stride;
-  const unsigned lds = (unsigned)(mlp_lds_bytes(sh.maxdim) +
-                                  (((sdim + 3) & ~3) + 64 + 8) * 4);
+  unsigned lds;
+  int rc = loco_prepare(&sh, &la, dims_host, ndims, row_stride, ob_clip, sdim, adim,
+                        goal_flag, terminate, noiseless_from, bins, leak, ctrl,
+                        alive_bonus, fall_thr, dt, &lds);
+  if (rc) return rc;
+  LocoPtrs P = loco_ptrs(weights, obmean, obstd, ac_std_dev, seed_dev, s_glob, pos, goal,
+                         Am, Bm, b0, wv, wa, wy, wh, alive, rew_total, member_steps, behv,
+                         mo_sum, mo_sumsq);
   loco_step_kernel<<<dim3((unsigned)n_pop), dim3(256), lds, (hipStream_t)stream>>>(
-      (const uint16_t*)weights, (const float*)obmean, (const float*)obstd, sh, la,
-      (const float*)ac_std_dev, (const uint64_t*)seed_dev, (float*)s_glob, (float*)pos, (const float*)goal,
-      (const float*)Am, (const float*)Bm, (const float*)b0, (const float*)wv,
-      (const float*)wa, (const float*)wy, (const float*)wh, (float*)alive,
-      (float*)rew_total, (float*)member_steps, (float*)behv, (float*)mo_sum,
-      (float*)mo_sumsq);
+      sh, la, P, salt);
+  ES_CHECK_LAUNCH();
+  return 0;
+}
+
+extern "C" int es_loco_episode(const void* weights, const void* obmean, const void* obstd,
+                               const int32_t* dims_host, int32_t ndims, const void* seed_dev,
+                               int32_t n_steps, float ob_clip, const void* ac_std_dev,
+                               int64_t row_stride,
+                               void* s_glob, void* pos, const void* goal, const void* Am,
+                               const void* Bm, const void* b0, const void* wv,
+                               const void* wa, const void* wy, const void* wh, void* alive,
+                               void* rew_total, void* member_steps, void* behv,
+                               void* mo_sum, void* mo_sumsq, int32_t member_base,
+                               int32_t n_members, int32_t sdim, int32_t adim,
+                               int32_t goal_flag, int32_t terminate, int32_t noiseless_from,
+                               int32_t bins, float leak, float ctrl, float alive_bonus,
+                               float fall_thr, float dt, void* stream) {
+  MlpShape sh;
+  LocoArgs la;
+  unsigned lds;
+  int rc = loco_prepare(&sh, &la, dims_host, ndims, row_stride, ob_clip, sdim, adim,
+                        goal_flag, terminate, noiseless_from, bins, leak, ctrl,
+                        alive_bonus, fall_thr, dt, &lds);
+  if (rc) return rc;
+  LocoPtrs P = loco_ptrs(weights, obmean, obstd, ac_std_dev, seed_dev, s_glob, pos, goal,
+                         Am, Bm, b0, wv, wa, wy, wh, alive, rew_total, member_steps, behv,
+                         mo_sum, mo_sumsq);
+  loco_episode_kernel<<<dim3((unsigned)n_members), dim3(256), lds, (hipStream_t)stream>>>(
+      sh, la, P, member_base, n_steps);
   ES_CHECK_LAUNCH();
   return 0;
 }
