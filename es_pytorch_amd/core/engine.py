@@ -94,7 +94,10 @@ class GpuEngine:
         assert ppg % comm.size == 0 and (ppg / comm.size) % 2 == 0, \
             "policies_per_gen must split into antithetic pairs per rank"
         self.pairs = int(ppg // comm.size // 2)
-        self.B = 2 * self.pairs + 1  # [+pairs | -pairs | noiseless]
+        # episodes per perturbation (reference obj.py:56-63 averaging)
+        self.eps = max(1, int(cfg.general.get("eps_per_policy", 1) or 1))
+        self.M = 2 * self.pairs + 1      # members: [+pairs | -pairs | noiseless]
+        self.B = self.M * self.eps       # evaluation slots (env batch)
         assert env.batch == self.B, f"env batch {env.batch} != engine batch {self.B}"
 
         self.dims = policy._module.layer_dims()
@@ -115,8 +118,8 @@ class GpuEngine:
         # member blob rows padded to 16 B so the forward kernel's uint4
         # (8 x bf16) vector loads stay aligned for every member
         self.row_stride = (self.n + 7) // 8 * 8
-        self.weights = torch.empty((self.B, self.row_stride), dtype=torch.bfloat16, device=d)
-        self.offsets = torch.zeros(self.B, dtype=torch.int64, device=d)
+        self.weights = torch.empty((self.M, self.row_stride), dtype=torch.bfloat16, device=d)
+        self.offsets = torch.zeros(self.M, dtype=torch.int64, device=d)
         self.signs = torch.cat([torch.ones(self.pairs), -torch.ones(self.pairs),
                                 torch.zeros(1)]).to(d)
         self.grad = torch.empty(self.n, dtype=torch.float32, device=d)
@@ -177,7 +180,7 @@ class GpuEngine:
         std = float(self.policy.std)
         ops.check(ops.hip().es_pheno_bf16(
             self.weights.data_ptr(), self.theta.data_ptr(), self.nt.noise.data_ptr(),
-            self.offsets.data_ptr(), self.signs.data_ptr(), self.B, self.n,
+            self.offsets.data_ptr(), self.signs.data_ptr(), self.M, self.n,
             self.row_stride, std, self._stream()), "es_pheno_bf16")
 
     def _forward(self, obs: torch.Tensor, salt: int):
@@ -186,7 +189,7 @@ class GpuEngine:
             self.obmean.data_ptr(), self.obstd.data_ptr(),
             self.dims_arr.ctypes.data, len(self.dims_arr), self.seed_dev.data_ptr(),
             salt, self.B, float(self.policy._module.ob_clip), self.acstd_dev.data_ptr(),
-            self.row_stride, 1, self.B - 1, self.bins,
+            self.row_stride, 1, (self.M - 1) * self.eps, self.bins, self.eps,
             self.alow_dev.data_ptr() if self.alow_dev is not None else None,
             self.arange_dev.data_ptr() if self.arange_dev is not None else None,
             self._stream()), "es_mlp_fwd")
@@ -209,8 +212,8 @@ class GpuEngine:
             self.alive.data_ptr(), self.rew_total.data_ptr(),
             self.member_steps.data_ptr(), self.behv.data_ptr(),
             self.mo_sum.data_ptr(), self.mo_sumsq.data_ptr(),
-            self.B - 1, env.sdim, env.ac_dim, int(env.goal_conditioned),
-            int(env.terminate_on_fall), self.B - 1, self.bins,
+            (self.M - 1) * self.eps, env.sdim, env.ac_dim, int(env.goal_conditioned),
+            int(env.terminate_on_fall), (self.M - 1) * self.eps, self.bins, self.eps,
             float(env.leak), float(env.ctrl_cost), float(env.alive_bonus),
             float(env.fall_threshold), float(env.dt), self._stream()), "es_loco_step")
 
@@ -229,8 +232,8 @@ class GpuEngine:
             self.alive.data_ptr(), self.rew_total.data_ptr(),
             self.member_steps.data_ptr(), self.behv.data_ptr(),
             self.mo_sum.data_ptr(), self.mo_sumsq.data_ptr(),
-            self.B - 1, 1, env.sdim, env.ac_dim, int(env.goal_conditioned),
-            int(env.terminate_on_fall), 0, self.bins,
+            (self.M - 1) * self.eps, self.eps, env.sdim, env.ac_dim,
+            int(env.goal_conditioned), int(env.terminate_on_fall), 0, self.bins, self.eps,
             float(env.leak), float(env.ctrl_cost), float(env.alive_bonus),
             float(env.fall_threshold), float(env.dt), self._stream()),
             "es_loco_episode")
@@ -311,22 +314,34 @@ class GpuEngine:
         return (self.gen * 1000003 + self.comm.rank * 7919) & 0x7FFFFFFF
 
     # ------------------------------------------------------------- fitness
+    def _member_rewards(self) -> torch.Tensor:
+        """(M,) episode-averaged total reward per member (obj.py:56-63)."""
+        return self.rew_total.view(self.M, self.eps).mean(1)
+
+    def _member_behv(self) -> torch.Tensor:
+        """(M, 3) final behaviour; last episode's, like the reference's
+        multi-episode rollout (flagrun.py:105-140 keeps the last behv)."""
+        return self.behv.view(self.M, self.eps, 3)[:, -1]
+
     def _fitnesses(self) -> torch.Tensor:
-        """(B, O) fitness per member from accumulated rollout state."""
+        """(M, O) fitness per member from accumulated rollout state."""
+        rew = self._member_rewards()
+        behv = self._member_behv()
         if self.objective == "reward":
-            return self.rew_total.unsqueeze(1)
+            return rew.unsqueeze(1)
         if self.objective == "mean_reward":
-            return (self.rew_total / torch.clamp(self.member_steps, min=1.0)).unsqueeze(1)
+            steps = self.member_steps.view(self.M, self.eps).mean(1)
+            return (rew / torch.clamp(steps, min=1.0)).unsqueeze(1)
         if self.objective == "dist":
-            return self.behv[:, :2].norm(dim=1, keepdim=True)
+            return behv[:, :2].norm(dim=1, keepdim=True)
         if self.objective == "xdist":
-            return self.behv[:, :1]
+            return behv[:, :1]
         if self.objective in ("ns", "nsr"):
             assert self.archive is not None, "set engine.archive before ns/nsr generations"
-            nov = novelty_batch(self.behv[:, :2], self.archive, self.novelty_k)
+            nov = novelty_batch(behv[:, :2], self.archive, self.novelty_k)
             if self.objective == "ns":
                 return nov.unsqueeze(1)
-            return torch.stack([self.rew_total, nov], dim=1)
+            return torch.stack([rew, nov], dim=1)
         raise ValueError(f"unknown objective {self.objective!r}")
 
     # ---------------------------------------------------------------- step
@@ -344,7 +359,8 @@ class GpuEngine:
         self.acstd_dev.fill_(float(getattr(self.policy._module, "_action_std", 0.0)))
         chance = float(cfg.policy.get("save_obs_chance", 1.0))
         sm = (self.rs.random_sample(self.B) < chance).astype(np.float32)
-        sm[-1] = 0.0  # noiseless slot never feeds the gen obstat (reference es.py:48 is separate)
+        # noiseless slots never feed the gen obstat (reference es.py:48 is separate)
+        sm[(self.M - 1) * self.eps:] = 0.0
         self.save_mask.copy_(torch.from_numpy(sm).to(self.device))
 
         self._pheno()
@@ -373,7 +389,7 @@ class GpuEngine:
         all_rows = self.comm.allgather_rows(rows).cpu().numpy()
         # evaluated-episode steps only (the reference's es.py:79 counts the
         # pos+neg rollouts, not the noiseless eval)
-        local_steps = float(self.member_steps[:2 * self.pairs].sum().item())
+        local_steps = float(self.member_steps[:2 * self.pairs * self.eps].sum().item())
         steps = int(self.comm.allreduce_scalar(local_steps))
         t3 = time.perf_counter()
 
@@ -409,8 +425,8 @@ class GpuEngine:
                            float(self.ob_count.item()))
         gen_obstat.dist_inc(self.comm)
 
-        nl_rew = float(self.rew_total[-1].item())
-        nl_pos = self.behv[-1].cpu().numpy()
+        nl_rew = float(self._member_rewards()[-1].item())
+        nl_pos = self._member_behv()[-1].cpu().numpy()
         noiseless = _NoiselessResult(nl_rew, nl_pos, steps, self.env.observation_space.shape)
         if reporter is not None:
             reporter.log_gen(np.concatenate([pos, neg]), noiseless, self.policy, steps)

@@ -14,7 +14,7 @@ mlp_fwd_kernel(float* __restrict__ actions, const float* __restrict__ obs,
                const float* __restrict__ obstd, MlpShape sh, float ob_clip,
                const float* __restrict__ ac_std_dev,
                const uint64_t* __restrict__ seed_dev, uint64_t salt, int64_t row_stride,
-               int act_final, int noiseless_from, int bins,
+               int act_final, int noiseless_from, int bins, int eps,
                const float* __restrict__ alow, const float* __restrict__ arange) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   float* bufA = reinterpret_cast<float*>(smem);
@@ -22,7 +22,7 @@ mlp_fwd_kernel(float* __restrict__ actions, const float* __restrict__ obs,
   float* partial = bufB + sh.maxdim;
   const int b = blockIdx.x;
   const int tid = threadIdx.x;
-  const uint16_t* wb = weights + (int64_t)b * row_stride;
+  const uint16_t* wb = weights + (int64_t)(b / eps) * row_stride;
   const int D = sh.dims[0];
 
   for (int i = tid; i < D; i += blockDim.x) {
@@ -69,7 +69,8 @@ extern "C" int es_mlp_fwd(void* actions, const void* obs, const void* weights,
                           int32_t ndims, const void* seed_dev, uint64_t salt, int32_t n_pop,
                           float ob_clip, const void* ac_std_dev, int64_t row_stride,
                           int32_t act_final, int32_t noiseless_from, int32_t bins,
-                          const void* alow, const void* arange, void* stream) {
+                          int32_t eps, const void* alow, const void* arange,
+                          void* stream) {
   MlpShape sh;
   int rc = mlp_shape_init(&sh, dims_host, ndims, row_stride);
   if (rc) return rc;
@@ -78,7 +79,7 @@ extern "C" int es_mlp_fwd(void* actions, const void* obs, const void* weights,
       (float*)actions, (const float*)obs, (const uint16_t*)weights, (const float*)obmean,
       (const float*)obstd, sh, ob_clip, (const float*)ac_std_dev,
       (const uint64_t*)seed_dev, salt, row_stride, act_final, noiseless_from, bins,
-      (const float*)alow, (const float*)arange);
+      eps > 0 ? eps : 1, (const float*)alow, (const float*)arange);
   ES_CHECK_LAUNCH();
   return 0;
 }

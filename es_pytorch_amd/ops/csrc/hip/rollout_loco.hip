@@ -27,6 +27,7 @@ struct LocoArgs {
   int terminate;       // terminate_on_fall
   int noiseless_from;  // members >= this index get no action noise
   int bins;            // >1: K9 binned-action decode (FFBinned)
+  int eps;             // episodes per perturbation: slot b uses weights row b/eps
   float leak, ctrl, alive_bonus, fall_thr, dt, ob_clip;
   int64_t row_stride;
 };
@@ -62,7 +63,7 @@ __device__ __forceinline__ void loco_step_body(
   __syncthreads();
 
   // ---- policy forward ----------------------------------------------------
-  const uint16_t* wb = P.weights + (int64_t)b * la.row_stride;
+  const uint16_t* wb = P.weights + (int64_t)(b / la.eps) * la.row_stride;
   const float* aout = mlp_layers(wb, sh, bufA, bufB, partial, tid, nth, 1);
   const uint64_t seed = P.seed_dev ? (*P.seed_dev + salt) : salt;
   const float ac_std = P.ac_std_dev ? *P.ac_std_dev : 0.0f;
@@ -255,8 +256,8 @@ loco_episode_kernel(MlpShape sh, LocoArgs la, LocoPtrs P, int member_base, int n
 static int loco_prepare(MlpShape* sh, LocoArgs* la, const int32_t* dims_host, int32_t ndims,
                         int64_t row_stride, float ob_clip, int32_t sdim, int32_t adim,
                         int32_t goal_flag, int32_t terminate, int32_t noiseless_from,
-                        int32_t bins, float leak, float ctrl, float alive_bonus,
-                        float fall_thr, float dt, unsigned* lds) {
+                        int32_t bins, int32_t eps, float leak, float ctrl,
+                        float alive_bonus, float fall_thr, float dt, unsigned* lds) {
   int rc = mlp_shape_init(sh, dims_host, ndims, row_stride);
   if (rc) return rc;
   if (adim > 64 || sdim > ES_MAXDIM) return -103;
@@ -265,6 +266,7 @@ static int loco_prepare(MlpShape* sh, LocoArgs* la, const int32_t* dims_host, in
   if (sh->dims[sh->n_layers] != out_dim) return -105;
   la->S = sdim; la->A = adim; la->D = sh->dims[0]; la->goal = goal_flag;
   la->terminate = terminate; la->noiseless_from = noiseless_from; la->bins = bins;
+  la->eps = eps > 0 ? eps : 1;
   la->leak = leak; la->ctrl = ctrl; la->alive_bonus = alive_bonus; la->fall_thr = fall_thr;
   la->dt = dt; la->ob_clip = ob_clip; la->row_stride = row_stride;
   *lds = (unsigned)(mlp_lds_bytes(sh->maxdim) + (((sdim + 3) & ~3) + 64 + 8) * 4);
@@ -300,13 +302,13 @@ extern "C" int es_loco_step(const void* weights, const void* obmean, const void*
                             void* member_steps, void* behv, void* mo_sum, void* mo_sumsq,
                             int32_t n_pop, int32_t sdim, int32_t adim, int32_t goal_flag,
                             int32_t terminate, int32_t noiseless_from, int32_t bins,
-                            float leak, float ctrl, float alive_bonus, float fall_thr,
-                            float dt, void* stream) {
+                            int32_t eps, float leak, float ctrl, float alive_bonus,
+                            float fall_thr, float dt, void* stream) {
   MlpShape sh;
   LocoArgs la;
   unsigned lds;
   int rc = loco_prepare(&sh, &la, dims_host, ndims, row_stride, ob_clip, sdim, adim,
-                        goal_flag, terminate, noiseless_from, bins, leak, ctrl,
+                        goal_flag, terminate, noiseless_from, bins, eps, leak, ctrl,
                         alive_bonus, fall_thr, dt, &lds);
   if (rc) return rc;
   LocoPtrs P = loco_ptrs(weights, obmean, obstd, ac_std_dev, seed_dev, s_glob, pos, goal,
@@ -329,13 +331,13 @@ extern "C" int es_loco_episode(const void* weights, const void* obmean, const vo
                                void* mo_sum, void* mo_sumsq, int32_t member_base,
                                int32_t n_members, int32_t sdim, int32_t adim,
                                int32_t goal_flag, int32_t terminate, int32_t noiseless_from,
-                               int32_t bins, float leak, float ctrl, float alive_bonus,
-                               float fall_thr, float dt, void* stream) {
+                               int32_t bins, int32_t eps, float leak, float ctrl,
+                               float alive_bonus, float fall_thr, float dt, void* stream) {
   MlpShape sh;
   LocoArgs la;
   unsigned lds;
   int rc = loco_prepare(&sh, &la, dims_host, ndims, row_stride, ob_clip, sdim, adim,
-                        goal_flag, terminate, noiseless_from, bins, leak, ctrl,
+                        goal_flag, terminate, noiseless_from, bins, eps, leak, ctrl,
                         alive_bonus, fall_thr, dt, &lds);
   if (rc) return rc;
   LocoPtrs P = loco_ptrs(weights, obmean, obstd, ac_std_dev, seed_dev, s_glob, pos, goal,

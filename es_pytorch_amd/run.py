@@ -52,7 +52,8 @@ def build_run(cfg: AttrDict, objective: str = "reward", use_gpu: Optional[bool] 
 
     if use_gpu:
         pairs = int(cfg.general.policies_per_gen // comm.size // 2)
-        env = make_batched(cfg.env.name, 2 * pairs + 1, device, **env_kwargs)
+        eps = max(1, int(cfg.general.get("eps_per_policy", 1) or 1))
+        env = make_batched(cfg.env.name, (2 * pairs + 1) * eps, device, **env_kwargs)
     else:
         env = make(cfg.env.name, **env_kwargs)
         env.seed(my_seed)

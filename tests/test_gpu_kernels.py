@@ -411,3 +411,43 @@ def test_binned_policy_engine(dev):
         eng.step(ranker)
         fits[fused] = np.concatenate([ranker.fits_pos, ranker.fits_neg]).ravel()
     np.testing.assert_allclose(fits[False], fits[True], rtol=1e-3, atol=1e-2)
+
+
+def test_engine_eps_per_policy(dev):
+    """eps_per_policy > 1: episode-averaged member fitness (obj.py:56-63)."""
+    import numpy as np
+    from es_pytorch_amd.config import AttrDict
+    from es_pytorch_amd.core.engine import GpuEngine
+    from es_pytorch_amd.core.noisetable import NoiseTable
+    from es_pytorch_amd.core.policy import Policy
+    from es_pytorch_amd.envs import make_batched
+    from es_pytorch_amd.nn.nn import FeedForward
+    from es_pytorch_amd.nn.optimizers import Adam
+    from es_pytorch_amd.parallel.comm import Comm
+    from es_pytorch_amd.utils.rankers import CenteredRanker
+
+    torch.manual_seed(8)
+    comm = Comm(dev)
+    eps = 3
+    cfg = AttrDict({"env": {"name": "Hopper-v3", "max_steps": 25},
+                    "noise": {"tbl_size": 500_000, "std": 0.05},
+                    "policy": {"layer_sizes": [32], "ac_std": 0.0, "l2coeff": 0.005,
+                               "lr": 0.01, "ob_clip": 5, "save_obs_chance": 1.0},
+                    "general": {"policies_per_gen": 8, "batch_size": 500, "seed": 1,
+                                "eps_per_policy": eps}})
+    M = 9
+    env = make_batched("Hopper-v3", M * eps, dev, max_steps=25, terminate_on_fall=False)
+    nn = FeedForward([32], torch.nn.Tanh(), env, 0.0, 5)
+    policy = Policy(nn, 0.05, Adam(len(Policy.get_flat(nn)), 0.01))
+    nt = NoiseTable.create_shared(comm, 500_000, len(policy), seed=4, device=dev)
+    rs = np.random.RandomState(17)
+    eng = GpuEngine(cfg, comm, policy, nt, env, rs, use_graph=False)
+    assert eng.eps == eps and eng.B == M * eps and eng.M == M
+    ranker = CenteredRanker()
+    eng.step(ranker)
+    # episode-averaged fitness equals the mean of the member's slot rewards
+    rt = eng.rew_total.view(M, eps).cpu().numpy()
+    got = np.concatenate([ranker.fits_pos, ranker.fits_neg]).ravel()
+    np.testing.assert_allclose(got, rt[:M - 1].mean(1), rtol=1e-5)
+    # slot rewards within a member differ (different env inits) -> real averaging
+    assert np.abs(rt[:, 0] - rt[:, 1]).max() > 1e-4
