@@ -85,7 +85,7 @@ def test_mlp_fwd_kernel(dev):
     ops.check(ops.hip().es_mlp_fwd(actions.data_ptr(), obs.data_ptr(), weights.data_ptr(),
                                    obmean.data_ptr(), obstd.data_ptr(),
                                    dims_arr.ctypes.data, len(dims), None, 0, B,
-                                   5.0, None, stride, 1, B, 0, None, None, _stream(dev)), "mlp_fwd")
+                                   5.0, None, stride, 1, B, 0, 1, None, None, _stream(dev)), "mlp_fwd")
     torch.cuda.synchronize()
     ref = _torch_mlp_ref(obs, weights, dims, obmean, obstd, 5.0)
     assert torch.allclose(actions, ref, atol=2e-2, rtol=2e-2), \
@@ -109,7 +109,7 @@ def test_mlp_fwd_odd_output_dim(dev):
     ops.check(ops.hip().es_mlp_fwd(actions.data_ptr(), obs.data_ptr(), weights.data_ptr(),
                                    obmean.data_ptr(), obstd.data_ptr(),
                                    dims_arr.ctypes.data, len(dims), None, 0, B,
-                                   5.0, None, stride, 1, B, 0, None, None, _stream(dev)), "mlp_fwd")
+                                   5.0, None, stride, 1, B, 0, 1, None, None, _stream(dev)), "mlp_fwd")
     torch.cuda.synchronize()
     ref = _torch_mlp_ref(obs, weights, dims, obmean, obstd, 5.0)
     assert torch.allclose(actions, ref, atol=2e-2, rtol=2e-2)
@@ -136,7 +136,7 @@ def test_mlp_fwd_action_noise_statistics(dev):
                                        obmean.data_ptr(), obstd.data_ptr(),
                                        dims_arr.ctypes.data, len(dims), seed.data_ptr(),
                                        salt, B, 5.0, acstd.data_ptr(), stride, 1, B,
-                                       0, None, None, _stream(dev)), "mlp_fwd")
+                                       0, 1, None, None, _stream(dev)), "mlp_fwd")
     torch.cuda.synchronize()
     assert torch.equal(a1, a2)  # same (seed, salt) -> same noise
     noise = a1.flatten()
@@ -147,7 +147,7 @@ def test_mlp_fwd_action_noise_statistics(dev):
                                    obmean.data_ptr(), obstd.data_ptr(),
                                    dims_arr.ctypes.data, len(dims), seed.data_ptr(),
                                    6, B, 5.0, acstd.data_ptr(), stride, 1, B,
-                                   0, None, None, _stream(dev)), "mlp_fwd")
+                                   0, 1, None, None, _stream(dev)), "mlp_fwd")
     torch.cuda.synchronize()
     assert not torch.equal(a1, a3)  # different salt -> different noise
 
