@@ -85,12 +85,20 @@ __device__ __forceinline__ void loco_step_body(
   }
   __syncthreads();
 
-  // ---- dynamics: pre = s A + a B + b0 ; s' = (1-leak) s + leak tanh(pre) --
+  // ---- dynamics + fused epilogue ----------------------------------------
+  // The epilogue that materializes s' also accumulates the reward/behaviour
+  // reduction partials AND the per-member obs statistics in the same pass
+  // (one s'-sweep instead of three, and wave shuffles replace the 8-barrier
+  // LDS reduction tree).
+  const float w_alive = P.alive[b];  // pre-step alive, used as obstat weight
+  float p0 = 0, p1 = 0, p2 = 0, p3 = 0;
+  float* ms = P.mo_sum + (int64_t)b * la.D;
+  float* mq = P.mo_sumsq + (int64_t)b * la.D;
   {
     const bool quad = (S % 4 == 0);
+    const int OCT = S >> 2;
+    const int PART = quad ? nth / OCT : 0;
     if (quad) {
-      const int OCT = S >> 2;
-      const int PART = nth / OCT;
       const int oi = tid % OCT, ip = tid / OCT;
       float acc[4] = {0, 0, 0, 0};
       if (ip < PART) {
@@ -129,61 +137,60 @@ __device__ __forceinline__ void loco_step_body(
         for (int q = 0; q < 4; ++q) partial[(ip * OCT + oi) * 4 + q] = acc[q];
       }
       __syncthreads();
-      for (int o = tid; o < S; o += nth) {
-        float p = P.b0[o];
+    }
+    for (int o = tid; o < S; o += nth) {
+      float p = P.b0[o];
+      if (quad) {
         const int oo = o >> 2, j = o & 3;
         for (int pp = 0; pp < PART; ++pp) p += partial[(pp * OCT + oo) * 4 + j];
-        for (int k = 0; k < A; ++k) p = fmaf(abuf[k], P.Bm[(int64_t)k * S + o], p);
-        const float sn = (1.0f - la.leak) * raws[o] + la.leak * tanhf(p);
-        bufA[o] = sn;
-        sb[o] = sn;
-      }
-    } else {
-      for (int o = tid; o < S; o += nth) {
-        float p = P.b0[o];
+      } else {
         for (int i = 0; i < S; ++i) p = fmaf(raws[i], P.Am[(int64_t)i * S + o], p);
-        for (int k = 0; k < A; ++k) p = fmaf(abuf[k], P.Bm[(int64_t)k * S + o], p);
-        const float sn = (1.0f - la.leak) * raws[o] + la.leak * tanhf(p);
-        bufA[o] = sn;
-        sb[o] = sn;
       }
-    }
-  }
-  __syncthreads();
-
-  // ---- block reductions: vfwd, vy, h, sum(a^2) ---------------------------
-  {
-    float p0 = 0, p1 = 0, p2 = 0, p3 = 0;
-    for (int i = tid; i < S; i += nth) {
-      const float sn = bufA[i];
-      p0 = fmaf(sn, P.wv[i], p0);
-      p1 = fmaf(sn, P.wy[i], p1);
-      p2 = fmaf(sn, P.wh[i], p2);
+      for (int k = 0; k < A; ++k) p = fmaf(abuf[k], P.Bm[(int64_t)k * S + o], p);
+      const float sn = (1.0f - la.leak) * raws[o] + la.leak * tanhf(p);
+      sb[o] = sn;
+      p0 = fmaf(sn, P.wv[o], p0);
+      p1 = fmaf(sn, P.wy[o], p1);
+      p2 = fmaf(sn, P.wh[o], p2);
+      if (w_alive > 0.0f) {
+        ms[o] += sn;
+        mq[o] += sn * sn;
+      }
     }
     for (int j = tid; j < A; j += nth) {
       p0 = fmaf(0.5f * abuf[j], P.wa[j], p0);
       p3 = fmaf(abuf[j], abuf[j], p3);
     }
-    partial[tid] = p0;
-    partial[256 + tid] = p1;
-    partial[512 + tid] = p2;
-    partial[768 + tid] = p3;
-    __syncthreads();
-    for (int off = nth >> 1; off > 0; off >>= 1) {
-      if (tid < off) {
-        partial[tid] += partial[tid + off];
-        partial[256 + tid] += partial[256 + tid + off];
-        partial[512 + tid] += partial[512 + tid + off];
-        partial[768 + tid] += partial[768 + tid + off];
-      }
-      __syncthreads();
-    }
   }
+  // wave-level shuffle reduction, then one cross-wave pass through LDS
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) {
+    p0 += __shfl_down(p0, off);
+    p1 += __shfl_down(p1, off);
+    p2 += __shfl_down(p2, off);
+    p3 += __shfl_down(p3, off);
+  }
+  __syncthreads();  // partial[] reuse after the dynamics matvec
+  if ((tid & 63) == 0) {
+    const int wid = tid >> 6;
+    partial[wid * 4 + 0] = p0;
+    partial[wid * 4 + 1] = p1;
+    partial[wid * 4 + 2] = p2;
+    partial[wid * 4 + 3] = p3;
+  }
+  __syncthreads();
 
   // ---- scalar bookkeeping (thread 0) -------------------------------------
   if (tid == 0) {
-    const float vfwd = partial[0], vy = partial[256], h = partial[512], asq = partial[768];
-    const float alive_old = P.alive[b];
+    const int nw = nth >> 6;
+    float vfwd = 0, vy = 0, h = 0, asq = 0;
+    for (int wdx = 0; wdx < nw; ++wdx) {
+      vfwd += partial[wdx * 4 + 0];
+      vy += partial[wdx * 4 + 1];
+      h += partial[wdx * 4 + 2];
+      asq += partial[wdx * 4 + 3];
+    }
+    const float alive_old = w_alive;
     float rew;
     if (la.goal) {
       const float rx = P.goal[(int64_t)b * 2 + 0] - P.pos[(int64_t)b * 3 + 0];
@@ -207,25 +214,14 @@ __device__ __forceinline__ void loco_step_body(
       P.behv[(int64_t)b * 3 + 2] = h;
     }
     P.alive[b] = alive_old * (1.0f - done);
-    sc[4] = alive_old;
   }
   __syncthreads();
 
-  // ---- per-member obs statistics (post-step obs, alive-weighted) ---------
-  const float w = sc[4];
-  if (w > 0.0f) {
-    float* ms = P.mo_sum + (int64_t)b * la.D;
-    float* mq = P.mo_sumsq + (int64_t)b * la.D;
-    for (int i = tid; i < S; i += nth) {
-      const float o = bufA[i];
-      ms[i] += o;
-      mq[i] += o * o;
-    }
-    if (la.goal && tid < 2) {
-      const float rel = (P.goal[(int64_t)b * 2 + tid] - P.pos[(int64_t)b * 3 + tid]) * 0.1f;
-      ms[S + tid] += rel;
-      mq[S + tid] += rel * rel;
-    }
+  // goal-relative obs dims of the per-member obs statistics (needs new pos)
+  if (la.goal && w_alive > 0.0f && tid < 2) {
+    const float rel = (P.goal[(int64_t)b * 2 + tid] - P.pos[(int64_t)b * 3 + tid]) * 0.1f;
+    ms[S + tid] += rel;
+    mq[S + tid] += rel * rel;
   }
   __syncthreads();  // LDS reuse safety for the episode kernel's next step
 }
