@@ -77,7 +77,12 @@ class GpuEngine:
     def __init__(self, cfg, comm: Comm, policy: Policy, nt: NoiseTable, env: BatchedEnv,
                  rs: np.random.RandomState, objective: str = "reward",
                  use_graph: bool = True, novelty_k: int = 10,
-                 fused: Optional[bool] = None):
+                 fused: Optional[bool] = None, rollout_mode: str = "step"):
+        # rollout_mode: "step" = one kernel per env step for the population
+        # (graph-replayed); "episode" = ONE kernel per generation, each block
+        # runs its member's whole episode (members are mutually independent,
+        # so no per-step rendezvous is required)
+        self.rollout_mode = rollout_mode
         self.cfg = cfg
         self.comm = comm
         self.policy = policy
@@ -217,8 +222,8 @@ class GpuEngine:
             float(env.leak), float(env.ctrl_cost), float(env.alive_bonus),
             float(env.fall_threshold), float(env.dt), self._stream()), "es_loco_step")
 
-    def _loco_noiseless_episode(self):
-        """Whole noiseless episode (member B-1) in one kernel launch."""
+    def _loco_episode(self, member_base: int, n_members: int, noiseless_from: int):
+        """Whole episodes for [member_base, member_base+n_members) slots."""
         env = self.env
         goal_ptr = env.goal.data_ptr() if env.goal_conditioned else None
         ops.check(ops.hip().es_loco_episode(
@@ -232,11 +237,15 @@ class GpuEngine:
             self.alive.data_ptr(), self.rew_total.data_ptr(),
             self.member_steps.data_ptr(), self.behv.data_ptr(),
             self.mo_sum.data_ptr(), self.mo_sumsq.data_ptr(),
-            (self.M - 1) * self.eps, self.eps, env.sdim, env.ac_dim,
-            int(env.goal_conditioned), int(env.terminate_on_fall), 0, self.bins, self.eps,
+            member_base, n_members, env.sdim, env.ac_dim,
+            int(env.goal_conditioned), int(env.terminate_on_fall), noiseless_from,
+            self.bins, self.eps,
             float(env.leak), float(env.ctrl_cost), float(env.alive_bonus),
             float(env.fall_threshold), float(env.dt), self._stream()),
             "es_loco_episode")
+
+    def _loco_noiseless_episode(self):
+        self._loco_episode((self.M - 1) * self.eps, self.eps, 0)
 
     # ------------------------------------------------------------- rollout
     def _step_body(self, t: int):
@@ -258,6 +267,11 @@ class GpuEngine:
         self.obs_buf.copy_(ob)
 
     def _rollout_body(self):
+        if self.fused and self.rollout_mode == "episode":
+            # whole generation in one launch: every block runs its member's
+            # full episode, blocks drift freely (no per-step rendezvous)
+            self._loco_episode(0, self.B, (self.M - 1) * self.eps)
+            return
         if self.fused:
             # noiseless member: one whole-episode kernel on a side stream,
             # concurrent with the per-step population kernels

@@ -25,6 +25,7 @@ def main():
     p.add_argument("--layers", type=int, nargs="*", default=[256, 256])
     p.add_argument("--env", type=str, default="Humanoid-v2")
     p.add_argument("--graph", action="store_true")
+    p.add_argument("--episode", action="store_true", help="whole-episode rollout mode")
     args = p.parse_args()
 
     from es_pytorch_amd.config import AttrDict
@@ -54,7 +55,8 @@ def main():
     nt = NoiseTable.create_shared(comm, cfg.noise.tbl_size, len(policy), seed=5,
                                   device=dev)
     rs = np.random.RandomState(0)
-    eng = GpuEngine(cfg, comm, policy, nt, env, rs, use_graph=args.graph)
+    eng = GpuEngine(cfg, comm, policy, nt, env, rs, use_graph=args.graph,
+                    rollout_mode="episode" if args.episode else "step")
     ranker = CenteredRanker()
     eng.step(ranker)  # warmup
 
