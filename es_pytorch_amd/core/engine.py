@@ -149,7 +149,6 @@ class GpuEngine:
         self.actions = torch.empty((self.B, env.ac_dim), dtype=torch.float32, device=d)
         self.alive = torch.ones(self.B, dtype=torch.float32, device=d)
         self.rew_total = torch.zeros(self.B, dtype=torch.float32, device=d)
-        self.steps_total = torch.zeros((), dtype=torch.float32, device=d)
         self.member_steps = torch.zeros(self.B, dtype=torch.float32, device=d)
         self.behv = torch.zeros((self.B, 3), dtype=torch.float32, device=d)
         self.save_mask = torch.zeros(self.B, dtype=torch.float32, device=d)
@@ -253,7 +252,6 @@ class GpuEngine:
         ob, rew, done = self.env.step(a)
         alive = self.alive
         self.rew_total.add_(rew * alive)
-        self.steps_total.add_(alive.sum())
         self.member_steps.add_(alive)
         am = alive.bool().unsqueeze(1)
         self.behv.copy_(torch.where(am, self.env.positions, self.behv))
@@ -312,7 +310,6 @@ class GpuEngine:
     def _reset_rollout_state(self):
         self.alive.fill_(1.0)
         self.rew_total.zero_()
-        self.steps_total.zero_()
         self.member_steps.zero_()
         self.behv.zero_()
         self.ob_sum.zero_()
