@@ -36,6 +36,9 @@ def main():
     p.add_argument("--pop-per-gpu", type=int, default=1280)
     p.add_argument("--max-steps", type=int, default=1000, help="env steps per episode")
     p.add_argument("--env", type=str, default="Humanoid-v2")
+    p.add_argument("--objective", type=str, default="reward",
+                   choices=["reward", "nsr"],
+                   help="nsr = NSR-A style [reward, novelty] 2-objective ES")
     p.add_argument("--layers", type=int, nargs="*", default=[256, 256])
     p.add_argument("--tbl-size", type=int, default=250_000_000)
     p.add_argument("--no-graph", action="store_true")
@@ -74,8 +77,8 @@ def main():
     B = 2 * (ppg // world // 2) + 1
     # fixed-horizon synthetic rollouts: every counted env step is a fully
     # computed batched forward + dynamics step (stable, honest throughput)
-    env = make_batched(cfg.env.name, B, device, max_steps=args.max_steps,
-                       terminate_on_fall=False)
+    env_kwargs = dict(max_steps=args.max_steps, terminate_on_fall=False)
+    env = make_batched(cfg.env.name, B, device, **env_kwargs)
     nn = FeedForward(cfg.policy.layer_sizes, torch.nn.Tanh(), env, cfg.policy.ac_std,
                      cfg.policy.ob_clip)
     policy = Policy(nn, cfg.noise.std, Adam(len(Policy.get_flat(nn)), cfg.policy.lr))
@@ -89,11 +92,20 @@ def main():
                                   device=device)
 
     if use_cuda:
-        engine = GpuEngine(cfg, comm, policy, nt, env, rs, objective="reward",
+        engine = GpuEngine(cfg, comm, policy, nt, env, rs, objective=args.objective,
                            use_graph=not args.no_graph)
+        if args.objective == "nsr":
+            # seeded starter archive on device (NSR-A semantics: novelty vs
+            # the behaviour archive, grown per generation)
+            engine.archive = torch.randn(16, 2, dtype=torch.float64,
+                                         device=device) * 5.0
     else:
         engine = _CpuRefEngine(cfg, comm, policy, nt, env, rs)
-    ranker = CenteredRanker()
+    if args.objective == "nsr":
+        from es_pytorch_amd.utils.rankers import MultiObjectiveRanker
+        ranker = MultiObjectiveRanker(CenteredRanker(), 0.5)
+    else:
+        ranker = CenteredRanker()
 
     for _ in range(args.warmup):
         engine.step(ranker)
@@ -104,7 +116,11 @@ def main():
     t0 = time.perf_counter()
     steps_done = 0
     for _ in range(args.steps):
-        engine.step(ranker)
+        tr, _ = engine.step(ranker) if use_cuda else (engine.step(ranker), None)
+        if args.objective == "nsr" and use_cuda:
+            b = comm.broadcast_obj(list(tr.behaviour), src=0)
+            engine.archive = torch.cat([engine.archive, torch.tensor(
+                [b], dtype=torch.float64, device=device)])
         steps_done += engine.timings["env_steps"]
     comm.barrier()
     if use_cuda:
@@ -120,7 +136,9 @@ def main():
 
     if comm.rank == 0:
         out = {
-            "metric": "env-steps/sec (whole node), Humanoid-v2 ES",
+            "metric": ("env-steps/sec (whole node), HumanoidFlagrun NSR-A"
+                       if args.objective == "nsr" else
+                       "env-steps/sec (whole node), Humanoid-v2 ES"),
             "value": round(env_steps_per_sec, 1),
             "unit": "env-steps/s",
             "n_gpus": args.gpus,
@@ -139,6 +157,7 @@ def main():
                 "policies_per_gen": ppg,
                 "episode_len": args.max_steps,
                 "seq_len": args.max_steps,
+                "objective": args.objective,
                 "parallelism": f"dp{args.gpus}",
                 "gens_per_sec": round(gens_per_sec, 3),
                 "noise_table_elems": tbl,
