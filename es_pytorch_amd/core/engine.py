@@ -211,7 +211,7 @@ class GpuEngine:
             t + 1, float(self.policy._module.ob_clip), self.acstd_dev.data_ptr(),
             self.row_stride,
             env.s.data_ptr(), env.pos.data_ptr(), goal_ptr,
-            env.A.data_ptr(), env.B.data_ptr(), env.b0.data_ptr(),
+            env.A_bf16.data_ptr(), env.B.data_ptr(), env.b0.data_ptr(),
             env.wv.data_ptr(), env.wa.data_ptr(), env.wy.data_ptr(), env.wh.data_ptr(),
             self.alive.data_ptr(), self.rew_total.data_ptr(),
             self.member_steps.data_ptr(), self.behv.data_ptr(),
@@ -231,7 +231,7 @@ class GpuEngine:
             self.max_steps, float(self.policy._module.ob_clip),
             self.acstd_dev.data_ptr(), self.row_stride,
             env.s.data_ptr(), env.pos.data_ptr(), goal_ptr,
-            env.A.data_ptr(), env.B.data_ptr(), env.b0.data_ptr(),
+            env.A_bf16.data_ptr(), env.B.data_ptr(), env.b0.data_ptr(),
             env.wv.data_ptr(), env.wa.data_ptr(), env.wy.data_ptr(), env.wh.data_ptr(),
             self.alive.data_ptr(), self.rew_total.data_ptr(),
             self.member_steps.data_ptr(), self.behv.data_ptr(),

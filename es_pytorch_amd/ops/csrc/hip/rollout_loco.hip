@@ -37,7 +37,8 @@ struct LocoPtrs {
   const float *obmean, *obstd, *ac_std_dev;
   const uint64_t* seed_dev;
   float *s_glob, *pos;
-  const float *goal, *Am, *Bm, *b0, *wv, *wa, *wy, *wh;
+  const float *goal, *Bm, *b0, *wv, *wa, *wy, *wh;
+  const uint16_t* Am;  // bf16 state-transition matrix (S, S)
   float *alive, *rew_total, *member_steps, *behv, *mo_sum, *mo_sumsq;
 };
 
@@ -95,56 +96,53 @@ __device__ __forceinline__ void loco_step_body(
   float* ms = P.mo_sum + (int64_t)b * la.D;
   float* mq = P.mo_sumsq + (int64_t)b * la.D;
   {
-    const bool quad = (S % 4 == 0);
-    const int OCT = S >> 2;
-    const int PART = quad ? nth / OCT : 0;
-    if (quad) {
+    // bf16 A, octet-tiled like the policy layers (mlp_core.h scheme)
+    const bool oct8 = (S % 8 == 0);
+    const int OCT = S >> 3;
+    const int PART = oct8 ? nth / OCT : 0;
+    if (oct8) {
       const int oi = tid % OCT, ip = tid / OCT;
-      float acc[4] = {0, 0, 0, 0};
+      float acc[8];
+#pragma unroll
+      for (int q = 0; q < 8; ++q) acc[q] = 0.0f;
       if (ip < PART) {
-        const float* acol = P.Am + (oi << 2);
-        // software-pipelined register double-buffer (see mlp_core.h)
+        const uint16_t* acol = P.Am + (oi << 3);
         auto ld = [&](int i) {
-          return *reinterpret_cast<const float4*>(acol + (int64_t)i * S);
-        };
-        auto f4 = [&](const float4 w, const float xi) {
-          acc[0] = fmaf(w.x, xi, acc[0]);
-          acc[1] = fmaf(w.y, xi, acc[1]);
-          acc[2] = fmaf(w.z, xi, acc[2]);
-          acc[3] = fmaf(w.w, xi, acc[3]);
+          return *reinterpret_cast<const uint4*>(acol + (int64_t)i * S);
         };
         int i = ip;
         const int step4 = PART * 4;
         if (i + 3 * PART < S) {
-          float4 c0 = ld(i), c1 = ld(i + PART), c2 = ld(i + 2 * PART), c3 = ld(i + 3 * PART);
+          uint4 c0 = ld(i), c1 = ld(i + PART), c2 = ld(i + 2 * PART), c3 = ld(i + 3 * PART);
           for (; i + 7 * PART < S; i += step4) {
-            const float4 n0 = ld(i + 4 * PART), n1 = ld(i + 5 * PART),
-                         n2 = ld(i + 6 * PART), n3 = ld(i + 7 * PART);
-            f4(c0, raws[i]);
-            f4(c1, raws[i + PART]);
-            f4(c2, raws[i + 2 * PART]);
-            f4(c3, raws[i + 3 * PART]);
+            const uint4 n0 = ld(i + 4 * PART), n1 = ld(i + 5 * PART),
+                        n2 = ld(i + 6 * PART), n3 = ld(i + 7 * PART);
+            bf8_fma(c0, raws[i], acc);
+            bf8_fma(c1, raws[i + PART], acc);
+            bf8_fma(c2, raws[i + 2 * PART], acc);
+            bf8_fma(c3, raws[i + 3 * PART], acc);
             c0 = n0; c1 = n1; c2 = n2; c3 = n3;
           }
-          f4(c0, raws[i]);
-          f4(c1, raws[i + PART]);
-          f4(c2, raws[i + 2 * PART]);
-          f4(c3, raws[i + 3 * PART]);
+          bf8_fma(c0, raws[i], acc);
+          bf8_fma(c1, raws[i + PART], acc);
+          bf8_fma(c2, raws[i + 2 * PART], acc);
+          bf8_fma(c3, raws[i + 3 * PART], acc);
           i += step4;
         }
-        for (; i < S; i += PART) f4(ld(i), raws[i]);
+        for (; i < S; i += PART) bf8_fma(ld(i), raws[i], acc);
 #pragma unroll
-        for (int q = 0; q < 4; ++q) partial[(ip * OCT + oi) * 4 + q] = acc[q];
+        for (int q = 0; q < 8; ++q) partial[(ip * OCT + oi) * 8 + q] = acc[q];
       }
       __syncthreads();
     }
     for (int o = tid; o < S; o += nth) {
       float p = P.b0[o];
-      if (quad) {
-        const int oo = o >> 2, j = o & 3;
-        for (int pp = 0; pp < PART; ++pp) p += partial[(pp * OCT + oo) * 4 + j];
+      if (oct8) {
+        const int oo = o >> 3, j = o & 7;
+        for (int pp = 0; pp < PART; ++pp) p += partial[(pp * OCT + oo) * 8 + j];
       } else {
-        for (int i = 0; i < S; ++i) p = fmaf(raws[i], P.Am[(int64_t)i * S + o], p);
+        for (int i = 0; i < S; ++i)
+          p = fmaf(raws[i], bf2f(P.Am[(int64_t)i * S + o]), p);
       }
       for (int k = 0; k < A; ++k) p = fmaf(abuf[k], P.Bm[(int64_t)k * S + o], p);
       const float sn = (1.0f - la.leak) * raws[o] + la.leak * tanhf(p);
@@ -280,7 +278,7 @@ static LocoPtrs loco_ptrs(const void* weights, const void* obmean, const void* o
   P.obmean = (const float*)obmean; P.obstd = (const float*)obstd;
   P.ac_std_dev = (const float*)ac_std_dev; P.seed_dev = (const uint64_t*)seed_dev;
   P.s_glob = (float*)s_glob; P.pos = (float*)pos; P.goal = (const float*)goal;
-  P.Am = (const float*)Am; P.Bm = (const float*)Bm; P.b0 = (const float*)b0;
+  P.Am = (const uint16_t*)Am; P.Bm = (const float*)Bm; P.b0 = (const float*)b0;
   P.wv = (const float*)wv; P.wa = (const float*)wa; P.wy = (const float*)wy;
   P.wh = (const float*)wh; P.alive = (float*)alive; P.rew_total = (float*)rew_total;
   P.member_steps = (float*)member_steps; P.behv = (float*)behv;
