@@ -66,6 +66,9 @@ def main(cfg):
 
         tr, gen_obstat = step_any(cfg, comm, policy, nt, env, engine, fit_fn, rs,
                                   ranker, reporter)
+        if engine is not None:  # per-phase timers (rollout/collective/update)
+            reporter.log({k: round(v, 4) for k, v in engine.timings.items()
+                          if k.endswith("_s")})
 
         cfg.policy.ac_std = nn._action_std = nn._action_std * cfg.policy.get("ac_std_decay", 1)
         cfg.noise.std = policy.std = max(cfg.noise.std * cfg.noise.get("std_decay", 1),
