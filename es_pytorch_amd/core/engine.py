@@ -266,9 +266,16 @@ class GpuEngine:
 
     def _rollout_body(self):
         if self.fused and self.rollout_mode == "episode":
-            # whole generation in one launch: every block runs its member's
-            # full episode, blocks drift freely (no per-step rendezvous)
-            self._loco_episode(0, self.B, (self.M - 1) * self.eps)
+            # whole generation in one (+1 side) launch: every block runs its
+            # member's full episode, blocks drift freely (no per-step
+            # rendezvous); noiseless member on the side stream keeps the main
+            # grid an exact multiple of the CU slot count
+            main = torch.cuda.current_stream(self.device)
+            self._side.wait_stream(main)
+            with torch.cuda.stream(self._side):
+                self._loco_noiseless_episode()
+            self._loco_episode(0, (self.M - 1) * self.eps, (self.M - 1) * self.eps)
+            main.wait_stream(self._side)
             return
         if self.fused:
             # noiseless member: one whole-episode kernel on a side stream,
