@@ -451,3 +451,40 @@ def test_engine_eps_per_policy(dev):
     np.testing.assert_allclose(got, rt[:M - 1].mean(1), rtol=1e-5)
     # slot rewards within a member differ (different env inits) -> real averaging
     assert np.abs(rt[:, 0] - rt[:, 1]).max() > 1e-4
+
+
+def test_engine_generic_env_graph(dev):
+    """Generic (non-locomotion) torch-env engine path under hipGraph capture:
+    CartPole batched env + standalone forward kernel, graph == eager."""
+    import numpy as np
+    from es_pytorch_amd.config import AttrDict
+    from es_pytorch_amd.core.engine import GpuEngine
+    from es_pytorch_amd.core.noisetable import NoiseTable
+    from es_pytorch_amd.core.policy import Policy
+    from es_pytorch_amd.envs import make_batched
+    from es_pytorch_amd.nn.nn import FeedForward
+    from es_pytorch_amd.nn.optimizers import Adam
+    from es_pytorch_amd.parallel.comm import Comm
+    from es_pytorch_amd.utils.rankers import CenteredRanker
+
+    fits = {}
+    for use_graph in (False, True):
+        torch.manual_seed(12)
+        comm = Comm(dev)
+        cfg = AttrDict({"env": {"name": "CartPole-v1", "max_steps": 40},
+                        "noise": {"tbl_size": 300_000, "std": 0.05},
+                        "policy": {"layer_sizes": [16], "ac_std": 0.0, "l2coeff": 0.005,
+                                   "lr": 0.02, "ob_clip": 5, "save_obs_chance": 1.0},
+                        "general": {"policies_per_gen": 8, "batch_size": 100, "seed": 2}})
+        env = make_batched("CartPole-v1", 9, dev)
+        nn = FeedForward([16], torch.nn.Tanh(), env, 0.0, 5)
+        policy = Policy(nn, 0.05, Adam(len(Policy.get_flat(nn)), 0.02))
+        nt = NoiseTable.create_shared(comm, 300_000, len(policy), seed=9, device=dev)
+        rs = np.random.RandomState(31)
+        eng = GpuEngine(cfg, comm, policy, nt, env, rs, use_graph=use_graph)
+        assert not eng.fused  # CartPole takes the generic torch-env path
+        ranker = CenteredRanker()
+        eng.step(ranker)
+        fits[use_graph] = np.concatenate([ranker.fits_pos, ranker.fits_neg]).ravel()
+    np.testing.assert_allclose(fits[False], fits[True], rtol=1e-5, atol=1e-5)
+    assert np.abs(fits[True]).sum() > 0  # episodes produced reward
