@@ -108,6 +108,16 @@ class GpuEngine:
         self.dims = policy._module.layer_dims()
         # K9: FFBinned policies emit adim*bins logits decoded in-kernel
         self.bins = int(getattr(policy._module, "bins", 0))
+        # integrated-gaussian-action variants (reference nn.py:53-96): the net
+        # emits its own action std — mode 2 (first output) / 3 (second half).
+        # NOTE: the engine contract sizes the output layer adim+1 / 2*adim.
+        from es_pytorch_amd.nn.nn import FFIntegGausAction, FFIntegGausActionMulti
+        if isinstance(policy._module, FFIntegGausActionMulti):
+            self.act_mode = 3
+        elif isinstance(policy._module, FFIntegGausAction):
+            self.act_mode = 2
+        else:
+            self.act_mode = 0
         self.n = int(np.sum([I * O + O for I, O in zip(self.dims[:-1], self.dims[1:])]))
         assert self.n == len(policy), (self.n, len(policy))
         self.perm = forward_perm(self.dims).to(self.device)  # fwd idx -> flat idx
@@ -194,6 +204,7 @@ class GpuEngine:
             self.dims_arr.ctypes.data, len(self.dims_arr), self.seed_dev.data_ptr(),
             salt, self.B, float(self.policy._module.ob_clip), self.acstd_dev.data_ptr(),
             self.row_stride, 1, (self.M - 1) * self.eps, self.bins, self.eps,
+            self.act_mode,
             self.alow_dev.data_ptr() if self.alow_dev is not None else None,
             self.arange_dev.data_ptr() if self.arange_dev is not None else None,
             self._stream()), "es_mlp_fwd")
@@ -218,6 +229,7 @@ class GpuEngine:
             self.mo_sum.data_ptr(), self.mo_sumsq.data_ptr(),
             (self.M - 1) * self.eps, env.sdim, env.ac_dim, int(env.goal_conditioned),
             int(env.terminate_on_fall), (self.M - 1) * self.eps, self.bins, self.eps,
+            self.act_mode,
             float(env.leak), float(env.ctrl_cost), float(env.alive_bonus),
             float(env.fall_threshold), float(env.dt), self._stream()), "es_loco_step")
 
@@ -238,7 +250,7 @@ class GpuEngine:
             self.mo_sum.data_ptr(), self.mo_sumsq.data_ptr(),
             member_base, n_members, env.sdim, env.ac_dim,
             int(env.goal_conditioned), int(env.terminate_on_fall), noiseless_from,
-            self.bins, self.eps,
+            self.bins, self.eps, self.act_mode,
             float(env.leak), float(env.ctrl_cost), float(env.alive_bonus),
             float(env.fall_threshold), float(env.dt), self._stream()),
             "es_loco_episode")

@@ -110,19 +110,19 @@ def _bind_hip(lib):
     lib.es_noise_fill.argtypes = [p, i64, u64, u32, p]
     lib.es_pheno_bf16.argtypes = [p, p, p, p, p, i64, i64, i64, f32, p]
     lib.es_mlp_fwd.argtypes = [p, p, p, p, p, p, i32, p, u64, i32, f32, p, i64, i32,
-                               i32, i32, i32, p, p, p]
+                               i32, i32, i32, i32, p, p, p]
     lib.es_grad_gather.argtypes = [p, p, p, p, i64, i64, p]
     lib.es_adam_step.argtypes = [p, p, p, p, i64, f32, f32, f32, f32, f32, f32, p]
     lib.es_sgd_step.argtypes = [p, p, p, i64, f32, f32, f32, f32, p]
     lib.es_loco_step.argtypes = [p, p, p, p, i32, p, u64, f32, p, i64,
                                  p, p, p, p, p, p, p, p, p, p,
                                  p, p, p, p, p, p,
-                                 i32, i32, i32, i32, i32, i32, i32, i32,
+                                 i32, i32, i32, i32, i32, i32, i32, i32, i32,
                                  f32, f32, f32, f32, f32, p]
     lib.es_loco_episode.argtypes = [p, p, p, p, i32, p, i32, f32, p, i64,
                                     p, p, p, p, p, p, p, p, p, p,
                                     p, p, p, p, p, p,
-                                    i32, i32, i32, i32, i32, i32, i32, i32, i32,
+                                    i32, i32, i32, i32, i32, i32, i32, i32, i32, i32,
                                     f32, f32, f32, f32, f32, p]
     for fn in ["es_noise_fill", "es_pheno_bf16", "es_mlp_fwd", "es_grad_gather",
                "es_adam_step", "es_sgd_step", "es_loco_step", "es_loco_episode"]:

@@ -14,7 +14,7 @@ mlp_fwd_kernel(float* __restrict__ actions, const float* __restrict__ obs,
                const float* __restrict__ obstd, MlpShape sh, float ob_clip,
                const float* __restrict__ ac_std_dev,
                const uint64_t* __restrict__ seed_dev, uint64_t salt, int64_t row_stride,
-               int act_final, int noiseless_from, int bins, int eps,
+               int act_final, int noiseless_from, int bins, int eps, int act_mode,
                const float* __restrict__ alow, const float* __restrict__ arange) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   float* bufA = reinterpret_cast<float*>(smem);
@@ -35,6 +35,20 @@ mlp_fwd_kernel(float* __restrict__ actions, const float* __restrict__ obs,
 
   const int A = sh.dims[sh.n_layers];
   const uint64_t seed = seed_dev ? (*seed_dev + salt) : salt;
+  if (act_mode == 2 || act_mode == 3) {
+    // integrated gaussian actions (reference FFIntegGausAction(Multi),
+    // nn.py:53-96): the net emits its own action std — first output (mode 2)
+    // or the second half of the outputs (mode 3, abs)
+    const int adim = act_mode == 2 ? A - 1 : A / 2;
+    for (int d = tid; d < adim; d += blockDim.x) {
+      float a = act_mode == 2 ? out[1 + d] : out[d];
+      const float std_o = act_mode == 2 ? out[0] : fabsf(out[adim + d]);
+      if (b < noiseless_from && std_o != 0.0f)
+        a += std_o * es_actnoise(seed, (uint64_t)b * A + d);
+      actions[(int64_t)b * adim + d] = a;
+    }
+    return;
+  }
   if (bins > 1) {
     // K9 binned-action decode (reference FFBinned, nn.py:111-117): per-dim
     // argmax over bins, rescaled into [alow, alow+range]
@@ -69,8 +83,8 @@ extern "C" int es_mlp_fwd(void* actions, const void* obs, const void* weights,
                           int32_t ndims, const void* seed_dev, uint64_t salt, int32_t n_pop,
                           float ob_clip, const void* ac_std_dev, int64_t row_stride,
                           int32_t act_final, int32_t noiseless_from, int32_t bins,
-                          int32_t eps, const void* alow, const void* arange,
-                          void* stream) {
+                          int32_t eps, int32_t act_mode, const void* alow,
+                          const void* arange, void* stream) {
   MlpShape sh;
   int rc = mlp_shape_init(&sh, dims_host, ndims, row_stride);
   if (rc) return rc;
@@ -79,7 +93,7 @@ extern "C" int es_mlp_fwd(void* actions, const void* obs, const void* weights,
       (float*)actions, (const float*)obs, (const uint16_t*)weights, (const float*)obmean,
       (const float*)obstd, sh, ob_clip, (const float*)ac_std_dev,
       (const uint64_t*)seed_dev, salt, row_stride, act_final, noiseless_from, bins,
-      eps > 0 ? eps : 1, (const float*)alow, (const float*)arange);
+      eps > 0 ? eps : 1, act_mode, (const float*)alow, (const float*)arange);
   ES_CHECK_LAUNCH();
   return 0;
 }

@@ -27,6 +27,7 @@ struct LocoArgs {
   int terminate;       // terminate_on_fall
   int noiseless_from;  // members >= this index get no action noise
   int bins;            // >1: K9 binned-action decode (FFBinned)
+  int act_mode;        // 0 plain, 1 binned, 2 integ-gauss, 3 integ-gauss-multi
   int eps;             // episodes per perturbation: slot b uses weights row b/eps
   float leak, ctrl, alive_bonus, fall_thr, dt, ob_clip;
   int64_t row_stride;
@@ -68,7 +69,17 @@ __device__ __forceinline__ void loco_step_body(
   const float* aout = mlp_layers(wb, sh, bufA, bufB, partial, tid, nth, 1);
   const uint64_t seed = P.seed_dev ? (*P.seed_dev + salt) : salt;
   const float ac_std = P.ac_std_dev ? *P.ac_std_dev : 0.0f;
-  if (la.bins > 1) {
+  if (la.act_mode == 2 || la.act_mode == 3) {
+    // integrated gaussian actions (reference nn.py:53-96)
+    const int odim = sh.dims[sh.n_layers];
+    if (tid < A) {
+      float a = la.act_mode == 2 ? aout[1 + tid] : aout[tid];
+      const float std_o = la.act_mode == 2 ? aout[0] : fabsf(aout[A + tid]);
+      if (b < la.noiseless_from && std_o != 0.0f)
+        a += std_o * es_actnoise(seed, (uint64_t)b * odim + tid);
+      abuf[tid] = fclampf(a, -1.0f, 1.0f);
+    }
+  } else if (la.bins > 1) {
     // K9 binned decode (FFBinned): per-dim argmax over bins -> [-1, 1]
     if (tid < A) {
       const float* row = aout + tid * la.bins;
@@ -250,16 +261,21 @@ loco_episode_kernel(MlpShape sh, LocoArgs la, LocoPtrs P, int member_base, int n
 static int loco_prepare(MlpShape* sh, LocoArgs* la, const int32_t* dims_host, int32_t ndims,
                         int64_t row_stride, float ob_clip, int32_t sdim, int32_t adim,
                         int32_t goal_flag, int32_t terminate, int32_t noiseless_from,
-                        int32_t bins, int32_t eps, float leak, float ctrl,
-                        float alive_bonus, float fall_thr, float dt, unsigned* lds) {
+                        int32_t bins, int32_t eps, int32_t act_mode, float leak,
+                        float ctrl, float alive_bonus, float fall_thr, float dt,
+                        unsigned* lds) {
   int rc = mlp_shape_init(sh, dims_host, ndims, row_stride);
   if (rc) return rc;
   if (adim > 64 || sdim > ES_MAXDIM) return -103;
   if (sh->dims[0] != sdim + (goal_flag ? 2 : 0)) return -104;
-  const int out_dim = bins > 1 ? adim * bins : adim;
+  int out_dim = adim;
+  if (bins > 1) out_dim = adim * bins;
+  else if (act_mode == 2) out_dim = adim + 1;
+  else if (act_mode == 3) out_dim = adim * 2;
   if (sh->dims[sh->n_layers] != out_dim) return -105;
   la->S = sdim; la->A = adim; la->D = sh->dims[0]; la->goal = goal_flag;
   la->terminate = terminate; la->noiseless_from = noiseless_from; la->bins = bins;
+  la->act_mode = act_mode;
   la->eps = eps > 0 ? eps : 1;
   la->leak = leak; la->ctrl = ctrl; la->alive_bonus = alive_bonus; la->fall_thr = fall_thr;
   la->dt = dt; la->ob_clip = ob_clip; la->row_stride = row_stride;
@@ -296,14 +312,14 @@ extern "C" int es_loco_step(const void* weights, const void* obmean, const void*
                             void* member_steps, void* behv, void* mo_sum, void* mo_sumsq,
                             int32_t n_pop, int32_t sdim, int32_t adim, int32_t goal_flag,
                             int32_t terminate, int32_t noiseless_from, int32_t bins,
-                            int32_t eps, float leak, float ctrl, float alive_bonus,
-                            float fall_thr, float dt, void* stream) {
+                            int32_t eps, int32_t act_mode, float leak, float ctrl,
+                            float alive_bonus, float fall_thr, float dt, void* stream) {
   MlpShape sh;
   LocoArgs la;
   unsigned lds;
   int rc = loco_prepare(&sh, &la, dims_host, ndims, row_stride, ob_clip, sdim, adim,
-                        goal_flag, terminate, noiseless_from, bins, eps, leak, ctrl,
-                        alive_bonus, fall_thr, dt, &lds);
+                        goal_flag, terminate, noiseless_from, bins, eps, act_mode, leak,
+                        ctrl, alive_bonus, fall_thr, dt, &lds);
   if (rc) return rc;
   LocoPtrs P = loco_ptrs(weights, obmean, obstd, ac_std_dev, seed_dev, s_glob, pos, goal,
                          Am, Bm, b0, wv, wa, wy, wh, alive, rew_total, member_steps, behv,
@@ -325,14 +341,15 @@ extern "C" int es_loco_episode(const void* weights, const void* obmean, const vo
                                void* mo_sum, void* mo_sumsq, int32_t member_base,
                                int32_t n_members, int32_t sdim, int32_t adim,
                                int32_t goal_flag, int32_t terminate, int32_t noiseless_from,
-                               int32_t bins, int32_t eps, float leak, float ctrl,
-                               float alive_bonus, float fall_thr, float dt, void* stream) {
+                               int32_t bins, int32_t eps, int32_t act_mode, float leak,
+                               float ctrl, float alive_bonus, float fall_thr, float dt,
+                               void* stream) {
   MlpShape sh;
   LocoArgs la;
   unsigned lds;
   int rc = loco_prepare(&sh, &la, dims_host, ndims, row_stride, ob_clip, sdim, adim,
-                        goal_flag, terminate, noiseless_from, bins, eps, leak, ctrl,
-                        alive_bonus, fall_thr, dt, &lds);
+                        goal_flag, terminate, noiseless_from, bins, eps, act_mode, leak,
+                        ctrl, alive_bonus, fall_thr, dt, &lds);
   if (rc) return rc;
   LocoPtrs P = loco_ptrs(weights, obmean, obstd, ac_std_dev, seed_dev, s_glob, pos, goal,
                          Am, Bm, b0, wv, wa, wy, wh, alive, rew_total, member_steps, behv,

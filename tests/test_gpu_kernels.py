@@ -85,7 +85,7 @@ def test_mlp_fwd_kernel(dev):
     ops.check(ops.hip().es_mlp_fwd(actions.data_ptr(), obs.data_ptr(), weights.data_ptr(),
                                    obmean.data_ptr(), obstd.data_ptr(),
                                    dims_arr.ctypes.data, len(dims), None, 0, B,
-                                   5.0, None, stride, 1, B, 0, 1, None, None, _stream(dev)), "mlp_fwd")
+                                   5.0, None, stride, 1, B, 0, 1, 0, None, None, _stream(dev)), "mlp_fwd")
     torch.cuda.synchronize()
     ref = _torch_mlp_ref(obs, weights, dims, obmean, obstd, 5.0)
     assert torch.allclose(actions, ref, atol=2e-2, rtol=2e-2), \
@@ -109,7 +109,7 @@ def test_mlp_fwd_odd_output_dim(dev):
     ops.check(ops.hip().es_mlp_fwd(actions.data_ptr(), obs.data_ptr(), weights.data_ptr(),
                                    obmean.data_ptr(), obstd.data_ptr(),
                                    dims_arr.ctypes.data, len(dims), None, 0, B,
-                                   5.0, None, stride, 1, B, 0, 1, None, None, _stream(dev)), "mlp_fwd")
+                                   5.0, None, stride, 1, B, 0, 1, 0, None, None, _stream(dev)), "mlp_fwd")
     torch.cuda.synchronize()
     ref = _torch_mlp_ref(obs, weights, dims, obmean, obstd, 5.0)
     assert torch.allclose(actions, ref, atol=2e-2, rtol=2e-2)
@@ -136,7 +136,7 @@ def test_mlp_fwd_action_noise_statistics(dev):
                                        obmean.data_ptr(), obstd.data_ptr(),
                                        dims_arr.ctypes.data, len(dims), seed.data_ptr(),
                                        salt, B, 5.0, acstd.data_ptr(), stride, 1, B,
-                                       0, 1, None, None, _stream(dev)), "mlp_fwd")
+                                       0, 1, 0, None, None, _stream(dev)), "mlp_fwd")
     torch.cuda.synchronize()
     assert torch.equal(a1, a2)  # same (seed, salt) -> same noise
     noise = a1.flatten()
@@ -147,7 +147,7 @@ def test_mlp_fwd_action_noise_statistics(dev):
                                    obmean.data_ptr(), obstd.data_ptr(),
                                    dims_arr.ctypes.data, len(dims), seed.data_ptr(),
                                    6, B, 5.0, acstd.data_ptr(), stride, 1, B,
-                                   0, 1, None, None, _stream(dev)), "mlp_fwd")
+                                   0, 1, 0, None, None, _stream(dev)), "mlp_fwd")
     torch.cuda.synchronize()
     assert not torch.equal(a1, a3)  # different salt -> different noise
 
@@ -488,3 +488,45 @@ def test_engine_generic_env_graph(dev):
         fits[use_graph] = np.concatenate([ranker.fits_pos, ranker.fits_neg]).ravel()
     np.testing.assert_allclose(fits[False], fits[True], rtol=1e-5, atol=1e-5)
     assert np.abs(fits[True]).sum() > 0  # episodes produced reward
+
+
+def test_integ_gauss_engine(dev):
+    """FFIntegGausAction on the engine (act_mode 2): the net's own first
+    output is the action std; fused and generic kernel paths agree."""
+    import numpy as np
+    from es_pytorch_amd.config import AttrDict
+    from es_pytorch_amd.core.engine import GpuEngine
+    from es_pytorch_amd.core.noisetable import NoiseTable
+    from es_pytorch_amd.core.policy import Policy
+    from es_pytorch_amd.envs import make_batched
+    from es_pytorch_amd.nn.nn import FFIntegGausAction
+    from es_pytorch_amd.nn.optimizers import Adam
+    from es_pytorch_amd.parallel.comm import Comm
+    from es_pytorch_amd.spaces import Box
+    from es_pytorch_amd.utils.rankers import CenteredRanker
+
+    fits = {}
+    for fused in (False, True):
+        torch.manual_seed(14)
+        comm = Comm(dev)
+        cfg = AttrDict({"env": {"name": "Hopper-v3", "max_steps": 25},
+                        "noise": {"tbl_size": 400_000, "std": 0.05},
+                        "policy": {"layer_sizes": [32], "ac_std": 0.0, "l2coeff": 0.005,
+                                   "lr": 0.01, "ob_clip": 5, "save_obs_chance": 1.0},
+                        "general": {"policies_per_gen": 8, "batch_size": 100, "seed": 2}})
+        env = make_batched("Hopper-v3", 9, dev, max_steps=25, terminate_on_fall=False)
+
+        class _View:  # engine contract: output layer = adim + 1
+            observation_space = env.observation_space
+            action_space = Box(-1.0, 1.0, (env.ac_dim + 1,))
+
+        nn = FFIntegGausAction([32], torch.nn.Tanh(), _View, ac_std=0.0, ob_clip=5)
+        policy = Policy(nn, 0.05, Adam(len(Policy.get_flat(nn)), 0.01))
+        nt = NoiseTable.create_shared(comm, 400_000, len(policy), seed=8, device=dev)
+        rs = np.random.RandomState(23)
+        eng = GpuEngine(cfg, comm, policy, nt, env, rs, use_graph=False, fused=fused)
+        assert eng.act_mode == 2
+        ranker = CenteredRanker()
+        eng.step(ranker)
+        fits[fused] = np.concatenate([ranker.fits_pos, ranker.fits_neg]).ravel()
+    np.testing.assert_allclose(fits[False], fits[True], rtol=1e-3, atol=1e-2)
