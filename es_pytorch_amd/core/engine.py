@@ -83,6 +83,9 @@ class GpuEngine:
         # runs its member's whole episode (members are mutually independent,
         # so no per-step rendezvous is required)
         self.rollout_mode = rollout_mode
+        # steps per launch in "step" mode: k>1 runs k consecutive env steps
+        # per kernel launch (bounded block drift, fewer launch boundaries)
+        self.steps_per_launch = int(cfg.general.get("steps_per_launch", 1) or 1)
         self.cfg = cfg
         self.comm = comm
         self.policy = policy
@@ -233,14 +236,17 @@ class GpuEngine:
             float(env.leak), float(env.ctrl_cost), float(env.alive_bonus),
             float(env.fall_threshold), float(env.dt), self._stream()), "es_loco_step")
 
-    def _loco_episode(self, member_base: int, n_members: int, noiseless_from: int):
-        """Whole episodes for [member_base, member_base+n_members) slots."""
+    def _loco_episode(self, member_base: int, n_members: int, noiseless_from: int,
+                      n_steps: Optional[int] = None, salt_base: int = 0):
+        """n_steps consecutive env steps (default: the whole episode) for
+        [member_base, member_base+n_members) slots in one launch."""
         env = self.env
         goal_ptr = env.goal.data_ptr() if env.goal_conditioned else None
         ops.check(ops.hip().es_loco_episode(
             self.weights.data_ptr(), self.obmean.data_ptr(), self.obstd.data_ptr(),
             self.dims_arr.ctypes.data, len(self.dims_arr), self.seed_dev.data_ptr(),
-            self.max_steps, float(self.policy._module.ob_clip),
+            self.max_steps if n_steps is None else n_steps,
+            float(self.policy._module.ob_clip),
             self.acstd_dev.data_ptr(), self.row_stride,
             env.s.data_ptr(), env.pos.data_ptr(), goal_ptr,
             env.A_bf16.data_ptr(), env.B.data_ptr(), env.b0.data_ptr(),
@@ -248,7 +254,7 @@ class GpuEngine:
             self.alive.data_ptr(), self.rew_total.data_ptr(),
             self.member_steps.data_ptr(), self.behv.data_ptr(),
             self.mo_sum.data_ptr(), self.mo_sumsq.data_ptr(),
-            member_base, n_members, env.sdim, env.ac_dim,
+            member_base, n_members, salt_base, env.sdim, env.ac_dim,
             int(env.goal_conditioned), int(env.terminate_on_fall), noiseless_from,
             self.bins, self.eps, self.act_mode,
             float(env.leak), float(env.ctrl_cost), float(env.alive_bonus),
@@ -296,8 +302,15 @@ class GpuEngine:
             self._side.wait_stream(main)
             with torch.cuda.stream(self._side):
                 self._loco_noiseless_episode()
-            for t in range(self.max_steps):
-                self._loco_step(t)
+            k = self.steps_per_launch
+            if k > 1:
+                mm = (self.M - 1) * self.eps
+                for t0 in range(0, self.max_steps, k):
+                    self._loco_episode(0, mm, mm, n_steps=min(k, self.max_steps - t0),
+                                       salt_base=t0)
+            else:
+                for t in range(self.max_steps):
+                    self._loco_step(t)
             main.wait_stream(self._side)
         else:
             for t in range(self.max_steps):

@@ -251,11 +251,13 @@ loco_step_kernel(MlpShape sh, LocoArgs la, LocoPtrs P, uint64_t salt) {
 }
 
 __global__ void __launch_bounds__(256)
-loco_episode_kernel(MlpShape sh, LocoArgs la, LocoPtrs P, int member_base, int n_steps) {
+loco_episode_kernel(MlpShape sh, LocoArgs la, LocoPtrs P, int member_base, int n_steps,
+                    int salt_base) {
   ES_LOCO_CARVE();
   const int b = member_base + blockIdx.x;
   for (int t = 1; t <= n_steps; ++t)
-    loco_step_body(sh, la, P, b, (uint64_t)t, bufA, bufB, partial, raws, abuf, sc);
+    loco_step_body(sh, la, P, b, (uint64_t)(salt_base + t), bufA, bufB, partial, raws,
+                   abuf, sc);
 }
 
 static int loco_prepare(MlpShape* sh, LocoArgs* la, const int32_t* dims_host, int32_t ndims,
@@ -339,7 +341,8 @@ extern "C" int es_loco_episode(const void* weights, const void* obmean, const vo
                                const void* wa, const void* wy, const void* wh, void* alive,
                                void* rew_total, void* member_steps, void* behv,
                                void* mo_sum, void* mo_sumsq, int32_t member_base,
-                               int32_t n_members, int32_t sdim, int32_t adim,
+                               int32_t n_members, int32_t salt_base, int32_t sdim,
+                               int32_t adim,
                                int32_t goal_flag, int32_t terminate, int32_t noiseless_from,
                                int32_t bins, int32_t eps, int32_t act_mode, float leak,
                                float ctrl, float alive_bonus, float fall_thr, float dt,
@@ -355,7 +358,7 @@ extern "C" int es_loco_episode(const void* weights, const void* obmean, const vo
                          Am, Bm, b0, wv, wa, wy, wh, alive, rew_total, member_steps, behv,
                          mo_sum, mo_sumsq);
   loco_episode_kernel<<<dim3((unsigned)n_members), dim3(256), lds, (hipStream_t)stream>>>(
-      sh, la, P, member_base, n_steps);
+      sh, la, P, member_base, n_steps, salt_base);
   ES_CHECK_LAUNCH();
   return 0;
 }

@@ -26,6 +26,7 @@ def main():
     p.add_argument("--env", type=str, default="Humanoid-v2")
     p.add_argument("--graph", action="store_true")
     p.add_argument("--episode", action="store_true", help="whole-episode rollout mode")
+    p.add_argument("--chunk", type=int, default=1, help="env steps per launch")
     args = p.parse_args()
 
     from es_pytorch_amd.config import AttrDict
@@ -46,7 +47,8 @@ def main():
         "noise": {"tbl_size": 50_000_000, "std": 0.02},
         "policy": {"layer_sizes": list(args.layers), "ac_std": 0.01, "l2coeff": 0.005,
                    "lr": 0.01, "ob_clip": 5, "save_obs_chance": 0.01},
-        "general": {"policies_per_gen": args.pop, "batch_size": 500, "seed": 5},
+        "general": {"policies_per_gen": args.pop, "batch_size": 500, "seed": 5,
+                    "steps_per_launch": args.chunk},
     })
     B = args.pop + 1
     env = make_batched(args.env, B, dev, max_steps=args.steps, terminate_on_fall=False)
