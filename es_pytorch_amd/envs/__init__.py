@@ -11,9 +11,6 @@ explicit and dependency-free.
 from __future__ import annotations
 
 import re
-from typing import Optional
-
-import torch
 
 from es_pytorch_amd.envs.base import BatchedEnv, Env, SingleFromBatched  # noqa: F401
 from es_pytorch_amd.envs.classic import BatchedCartPole, BatchedPendulum

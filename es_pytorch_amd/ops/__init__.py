@@ -19,7 +19,6 @@ from __future__ import annotations
 import ctypes
 import os
 import subprocess
-import sys
 from typing import Optional
 
 _DIR = os.path.dirname(os.path.abspath(__file__))

@@ -13,8 +13,6 @@ same goal-progress reward semantics.
 import os
 import sys
 
-import numpy as np
-
 sys.path.insert(0, os.path.join(os.path.dirname(os.path.abspath(__file__)), ".."))
 
 from es_pytorch_amd.config import load_config, parse_args

@@ -30,7 +30,7 @@ from es_pytorch_amd.envs import make, make_batched
 from es_pytorch_amd.nn.nn import FeedForward
 from es_pytorch_amd.nn.optimizers import Adam
 from es_pytorch_amd.parallel.comm import init_comm, seed_all
-from es_pytorch_amd.rollout import NSRResult, run_model
+from es_pytorch_amd.rollout import NSRResult
 from es_pytorch_amd.run import episodic_fit_fn
 from es_pytorch_amd.utils.novelty import novelty, update_archive
 from es_pytorch_amd.utils.rankers import CenteredRanker, MultiObjectiveRanker
@@ -72,7 +72,8 @@ def main(cfg):
     # population of policies (reference nsra.py:97-101)
     if use_gpu:
         pairs = int(cfg.general.policies_per_gen // comm.size // 2)
-        env = make_batched(cfg.env.name, 2 * pairs + 1, device,
+        eps = max(1, int(cfg.general.get("eps_per_policy", 1) or 1))
+        env = make_batched(cfg.env.name, (2 * pairs + 1) * eps, device,
                            max_steps=cfg.env.max_steps)
     else:
         env = make(cfg.env.name, max_steps=cfg.env.max_steps)
