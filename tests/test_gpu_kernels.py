@@ -530,3 +530,62 @@ def test_integ_gauss_engine(dev):
         eng.step(ranker)
         fits[fused] = np.concatenate([ranker.fits_pos, ranker.fits_neg]).ravel()
     np.testing.assert_allclose(fits[False], fits[True], rtol=1e-3, atol=1e-2)
+
+
+def _engine_pair(dev, env_name, layers, pop, max_steps, seed=20, B=None):
+    import numpy as np
+    from es_pytorch_amd.config import AttrDict
+    from es_pytorch_amd.core.engine import GpuEngine
+    from es_pytorch_amd.core.noisetable import NoiseTable
+    from es_pytorch_amd.core.policy import Policy
+    from es_pytorch_amd.envs import make_batched
+    from es_pytorch_amd.nn.nn import FeedForward
+    from es_pytorch_amd.nn.optimizers import Adam
+    from es_pytorch_amd.parallel.comm import Comm
+    from es_pytorch_amd.utils.rankers import CenteredRanker
+
+    fits = {}
+    for fused in (False, True):
+        torch.manual_seed(seed)
+        comm = Comm(dev)
+        cfg = AttrDict({"env": {"name": env_name, "max_steps": max_steps},
+                        "noise": {"tbl_size": 3_000_000, "std": 0.02},
+                        "policy": {"layer_sizes": list(layers), "ac_std": 0.01,
+                                   "l2coeff": 0.005, "lr": 0.01, "ob_clip": 5,
+                                   "save_obs_chance": 1.0},
+                        "general": {"policies_per_gen": pop, "batch_size": 500,
+                                    "seed": 1}})
+        env = make_batched(env_name, pop + 1, dev, max_steps=max_steps,
+                           terminate_on_fall=False)
+        nn = FeedForward(list(layers), torch.nn.Tanh(), env, 0.01, 5)
+        policy = Policy(nn, 0.02, Adam(len(Policy.get_flat(nn)), 0.01))
+        nt = NoiseTable.create_shared(comm, 3_000_000, len(policy), seed=seed + 1,
+                                      device=dev)
+        rs = np.random.RandomState(seed + 2)
+        eng = GpuEngine(cfg, comm, policy, nt, env, rs, use_graph=False, fused=fused)
+        ranker = CenteredRanker()
+        eng.step(ranker)
+        fits[fused] = np.concatenate([ranker.fits_pos, ranker.fits_neg]).ravel()
+    return fits
+
+
+def test_wide_hidden_layer_parity(dev):
+    """1024-wide hidden layer: OCT=128/PART=2 vector-path shape."""
+    import numpy as np
+    fits = _engine_pair(dev, "Humanoid-v2", [1024], pop=4, max_steps=15)
+    np.testing.assert_allclose(fits[False], fits[True], rtol=1e-3, atol=1e-2)
+
+
+def test_odd_state_dim_env_parity(dev):
+    """Walker2d (S=17): scalar bf16 dynamics path (S % 8 != 0)."""
+    import numpy as np
+    fits = _engine_pair(dev, "Walker2d-v3", [64, 64], pop=8, max_steps=20, seed=30)
+    np.testing.assert_allclose(fits[False], fits[True], rtol=1e-3, atol=1e-2)
+
+
+def test_minimal_population(dev):
+    """pairs=1 (B=3): smallest legal engine shape incl. side-stream episode."""
+    import numpy as np
+    fits = _engine_pair(dev, "Hopper-v3", [16], pop=2, max_steps=10, seed=40)
+    assert fits[True].shape == (2,)
+    np.testing.assert_allclose(fits[False], fits[True], rtol=1e-3, atol=1e-2)
