@@ -589,3 +589,46 @@ def test_minimal_population(dev):
     fits = _engine_pair(dev, "Hopper-v3", [16], pop=2, max_steps=10, seed=40)
     assert fits[True].shape == (2,)
     np.testing.assert_allclose(fits[False], fits[True], rtol=1e-3, atol=1e-2)
+
+
+def test_engine_bitwise_determinism(dev):
+    """Same seeds + same config -> bitwise-identical parameters after 3 gens.
+
+    This is the multi-rank redundant-update contract (reference
+    utils.py:69-70 semantics): ranks rely on identical inputs producing
+    identical updates; any nondeterministic kernel would break it."""
+    import numpy as np
+    flats = []
+    for _ in range(2):
+        from es_pytorch_amd.config import AttrDict
+        from es_pytorch_amd.core.engine import GpuEngine
+        from es_pytorch_amd.core.noisetable import NoiseTable
+        from es_pytorch_amd.core.policy import Policy
+        from es_pytorch_amd.envs import make_batched
+        from es_pytorch_amd.nn.nn import FeedForward
+        from es_pytorch_amd.nn.optimizers import Adam
+        from es_pytorch_amd.parallel.comm import Comm
+        from es_pytorch_amd.utils.rankers import CenteredRanker
+
+        torch.manual_seed(50)
+        comm = Comm(dev)
+        cfg = AttrDict({"env": {"name": "Humanoid-v2", "max_steps": 30},
+                        "noise": {"tbl_size": 2_000_000, "std": 0.02},
+                        "policy": {"layer_sizes": [64, 64], "ac_std": 0.01,
+                                   "l2coeff": 0.005, "lr": 0.01, "ob_clip": 5,
+                                   "save_obs_chance": 0.5},
+                        "general": {"policies_per_gen": 16, "batch_size": 500,
+                                    "seed": 3}})
+        env = make_batched("Humanoid-v2", 17, dev, max_steps=30,
+                           terminate_on_fall=False)
+        nn = FeedForward([64, 64], torch.nn.Tanh(), env, 0.01, 5)
+        policy = Policy(nn, 0.02, Adam(len(Policy.get_flat(nn)), 0.01))
+        nt = NoiseTable.create_shared(comm, 2_000_000, len(policy), seed=7, device=dev)
+        rs = np.random.RandomState(51)
+        eng = GpuEngine(cfg, comm, policy, nt, env, rs, use_graph=True)
+        ranker = CenteredRanker()
+        for _ in range(3):
+            tr, ob = eng.step(ranker)
+            eng.update_obstat(ob)
+        flats.append(policy.flat_params.copy())
+    np.testing.assert_array_equal(flats[0], flats[1])
