@@ -87,7 +87,7 @@ class FFIntegGausAction(FeedForward):
 
     def forward(self, inp: Tensor, **kwargs) -> np.ndarray:
         rs: Optional[np.random.RandomState] = kwargs.get("rs")
-        out = self.model(self._normalize(inp).float()).numpy()
+        out = self.model(self._normalize(inp).float()).detach().numpy()
         action, action_std = out[1:], out[0]
         if action_std != 0 and rs is not None:
             action = action + rs.standard_normal(*action.shape) * action_std
@@ -99,7 +99,7 @@ class FFIntegGausActionMulti(FeedForward):
 
     def forward(self, inp: Tensor, **kwargs) -> np.ndarray:
         rs: Optional[np.random.RandomState] = kwargs.get("rs")
-        out = self.model(self._normalize(inp).float()).numpy()
+        out = self.model(self._normalize(inp).float()).detach().numpy()
         mid = len(out) // 2
         action, action_std = out[:mid], np.abs(out[mid:])
         if rs is not None:
