@@ -34,13 +34,41 @@ grad_gather_kernel(float* __restrict__ g, const float* __restrict__ table,
       s_off[i] = offsets[p0 + i];
     }
     __syncthreads();
-    for (int i = 0; i < tile; ++i) {
-      const float f = s_fit[i];
-      const float* row = table + s_off[i] + base;
+    // software-pipelined over population rows: row i+1's 4 loads issue
+    // before row i's FMAs (hipcc alone drains vmcnt per row — latency-bound,
+    // measured ~7x off the bandwidth bound)
+    const bool full = base + chunk <= n_params;
+    if (full && tile > 1) {
+      float c[ES_GRAD_ELEMS];
+      const float* row0 = table + s_off[0] + base;
 #pragma unroll
-      for (int k = 0; k < ES_GRAD_ELEMS; ++k) {
-        const int64_t t = base + k * blockDim.x + threadIdx.x;
-        if (t < n_params) acc[k] = fmaf(f, row[k * blockDim.x + threadIdx.x], acc[k]);
+      for (int k = 0; k < ES_GRAD_ELEMS; ++k)
+        c[k] = row0[k * blockDim.x + threadIdx.x];
+      for (int i = 0; i < tile - 1; ++i) {
+        const float* rown = table + s_off[i + 1] + base;
+        float nx[ES_GRAD_ELEMS];
+#pragma unroll
+        for (int k = 0; k < ES_GRAD_ELEMS; ++k)
+          nx[k] = rown[k * blockDim.x + threadIdx.x];
+        const float f = s_fit[i];
+#pragma unroll
+        for (int k = 0; k < ES_GRAD_ELEMS; ++k) {
+          acc[k] = fmaf(f, c[k], acc[k]);
+          c[k] = nx[k];
+        }
+      }
+      const float f = s_fit[tile - 1];
+#pragma unroll
+      for (int k = 0; k < ES_GRAD_ELEMS; ++k) acc[k] = fmaf(f, c[k], acc[k]);
+    } else {
+      for (int i = 0; i < tile; ++i) {
+        const float f = s_fit[i];
+        const float* row = table + s_off[i] + base;
+#pragma unroll
+        for (int k = 0; k < ES_GRAD_ELEMS; ++k) {
+          const int64_t t = base + k * blockDim.x + threadIdx.x;
+          if (t < n_params) acc[k] = fmaf(f, row[k * blockDim.x + threadIdx.x], acc[k]);
+        }
       }
     }
   }

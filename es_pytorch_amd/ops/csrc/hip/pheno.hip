@@ -19,10 +19,21 @@ __global__ void pheno_bf16_kernel(uint16_t* __restrict__ out, const float* __res
   const float s = signs[b] * std;
   const float* noise = table + offsets[b];
   uint16_t* ob = out + b * row_stride;
-  for (int64_t t = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; t < n_params;
-       t += (int64_t)gridDim.x * blockDim.x) {
-    ob[t] = f2bf(theta[t] + s * noise[t]);
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  int64_t t = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  // 4-deep pipeline: keep 8 loads in flight (scalar loads — the table slice
+  // has arbitrary 4 B alignment, so float4 is not available)
+  for (; t + 3 * stride < n_params; t += 4 * stride) {
+    const float th0 = theta[t], n0 = noise[t];
+    const float th1 = theta[t + stride], n1 = noise[t + stride];
+    const float th2 = theta[t + 2 * stride], n2 = noise[t + 2 * stride];
+    const float th3 = theta[t + 3 * stride], n3 = noise[t + 3 * stride];
+    ob[t] = f2bf(th0 + s * n0);
+    ob[t + stride] = f2bf(th1 + s * n1);
+    ob[t + 2 * stride] = f2bf(th2 + s * n2);
+    ob[t + 3 * stride] = f2bf(th3 + s * n3);
   }
+  for (; t < n_params; t += stride) ob[t] = f2bf(theta[t] + s * noise[t]);
 }
 
 extern "C" int es_pheno_bf16(void* out, const void* theta, const void* table,
