@@ -11,8 +11,8 @@
 #include "common.h"
 
 #define ES_GRAD_TILE 256
-#define ES_GRAD_ELEMS 4
 
+template <int ES_GRAD_ELEMS>
 __global__ void __launch_bounds__(256)
 grad_gather_kernel(float* __restrict__ g, const float* __restrict__ table,
                    const float* __restrict__ fits, const int64_t* __restrict__ offsets,
@@ -84,11 +84,26 @@ extern "C" int es_grad_gather(void* g, const void* table, const void* fits,
                               const void* offsets, int64_t n_pop, int64_t n_params,
                               void* stream) {
   const int threads = 256;
-  const int64_t chunk = (int64_t)threads * ES_GRAD_ELEMS;
+  // pick the per-thread element count so the grid fills the chip (a fixed
+  // ELEMS=4 gave only 163 blocks at n=167k — occupancy-starved)
+  int elems = 4;
+  while (elems > 1 && (n_params + (int64_t)threads * elems - 1) /
+                          ((int64_t)threads * elems) < 1024)
+    elems >>= 1;
+  const int64_t chunk = (int64_t)threads * elems;
   const int blocks = (int)((n_params + chunk - 1) / chunk);
-  grad_gather_kernel<<<dim3(blocks), dim3(threads), 0, (hipStream_t)stream>>>(
-      (float*)g, (const float*)table, (const float*)fits, (const int64_t*)offsets, n_pop,
-      n_params);
+  if (elems == 4)
+    grad_gather_kernel<4><<<dim3(blocks), dim3(threads), 0, (hipStream_t)stream>>>(
+        (float*)g, (const float*)table, (const float*)fits, (const int64_t*)offsets,
+        n_pop, n_params);
+  else if (elems == 2)
+    grad_gather_kernel<2><<<dim3(blocks), dim3(threads), 0, (hipStream_t)stream>>>(
+        (float*)g, (const float*)table, (const float*)fits, (const int64_t*)offsets,
+        n_pop, n_params);
+  else
+    grad_gather_kernel<1><<<dim3(blocks), dim3(threads), 0, (hipStream_t)stream>>>(
+        (float*)g, (const float*)table, (const float*)fits, (const int64_t*)offsets,
+        n_pop, n_params);
   ES_CHECK_LAUNCH();
   return 0;
 }
