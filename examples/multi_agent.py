@@ -78,6 +78,8 @@ def custom_test_params(comm, n: int, policies: List[Policy], nt: NoiseTable, env
 def main(cfg):
     comm = init_comm()
     rs, my_seed, _ = seed_all(comm, cfg.general.seed)
+    if torch.cuda.is_available():
+        return main_gpu(cfg, comm, rs)
     env = make_multiagent(cfg.env.name, max_steps=cfg.env.max_steps)
     env.seed(my_seed)
     reporter = StdoutReporter(comm)
@@ -108,6 +110,40 @@ def main(cfg):
                            cfg.general.batch_size, cfg.policy.l2coeff)
             reporter.log({f"agent{i} avg": float(np.mean(rows[:, 0]))})
         reporter.log({"steps": steps})
+        reporter.end_gen()
+        if gen % 10 == 0 and comm.rank == 0:
+            for i, p in enumerate(policies):
+                p.save(f"saved/{cfg.general.name}/agent{i}", str(gen))
+
+
+def main_gpu(cfg, comm, rs):
+    """GPU-batched co-evolution (beyond reference parity): all perturbations
+    of every policy play each other in batched env instances."""
+    from es_pytorch_amd.core.ma_engine import MultiAgentGpuEngine
+    from es_pytorch_amd.envs.multiagent import BatchedPursuitTag
+
+    reporter = StdoutReporter(comm)
+    pairs = int(cfg.general.policies_per_gen // comm.size // 2)
+    B = 2 * pairs + 1
+    env = BatchedPursuitTag(B, comm.device, max_steps=cfg.env.max_steps)
+    policies = []
+    for i in range(env.N_AGENTS):
+        nn = FeedForward(cfg.policy.layer_sizes, torch.nn.Tanh(), _AgentEnvView(env, i),
+                         cfg.policy.ac_std, cfg.policy.ob_clip)
+        policies.append(Policy(nn, cfg.noise.std, Adam(len(Policy.get_flat(nn)),
+                                                       cfg.policy.lr)))
+    nt = NoiseTable.create_shared(comm, cfg.noise.tbl_size, len(policies[0]),
+                                  reporter, cfg.general.seed, device=comm.device)
+    engine = MultiAgentGpuEngine(cfg, comm, policies, nt, env, rs)
+    for gen in range(cfg.general.gens):
+        reporter.start_gen()
+        rankers = [CenteredRanker() for _ in range(env.N_AGENTS)]
+        noiseless, obstats = engine.step(rankers)
+        engine.update_obstats(obstats)
+        for i, r in enumerate(rankers):
+            reporter.log({f"agent{i} avg": float(np.mean(r.fits[:, 0]))})
+            reporter.log({f"agent{i} noiseless": noiseless[i]})
+        reporter.log({"steps": engine.timings["env_steps"]})
         reporter.end_gen()
         if gen % 10 == 0 and comm.rank == 0:
             for i, p in enumerate(policies):
