@@ -56,6 +56,8 @@ def main():
     from es_pytorch_amd.utils.rankers import CenteredRanker
 
     use_cuda = torch.cuda.is_available() and not args.cpu
+    if args.objective == "nsr" and not use_cuda:
+        raise SystemExit("--objective nsr requires the GPU engine")
     comm = init_comm()
     world = comm.size
     if world > 1 and args.gpus != world:

@@ -115,3 +115,18 @@ def test_viz(tmp_path):
     fits_dir = tmp_path / "saved" / "CartPole-v1-t" / "fits"
     out2 = viz.graph_fits(str(fits_dir))
     assert os.path.exists(out2)
+
+
+def test_simple_example_torchrun_world2(tmp_path):
+    """simple_example under the torchrun launcher, 2 CPU ranks over gloo."""
+    cfg = _base(gens=1)
+    cfg_path = tmp_path / "cfg.json"
+    cfg_path.write_text(json.dumps(cfg))
+    cmd = [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+           "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+           "--master-port", "29523",
+           os.path.join(ROOT, "examples", "simple_example.py"), str(cfg_path)]
+    r = subprocess.run(cmd, capture_output=True, text=True, timeout=300,
+                       cwd=str(tmp_path), env=dict(os.environ, PYTHONPATH=ROOT))
+    assert r.returncode == 0, r.stderr[-3000:]
+    assert "avg fitness" in r.stdout
