@@ -682,3 +682,38 @@ def test_multi_agent_gpu_engine(dev):
         assert not np.array_equal(p.flat_params, f0)
         assert np.isfinite(p.flat_params).all()
     assert eng.timings["env_steps"] > 0
+
+
+def test_episode_rollout_mode_parity(dev):
+    """rollout_mode='episode' (whole episodes per launch) == per-step mode."""
+    import numpy as np
+    from es_pytorch_amd.config import AttrDict
+    from es_pytorch_amd.core.engine import GpuEngine
+    from es_pytorch_amd.core.noisetable import NoiseTable
+    from es_pytorch_amd.core.policy import Policy
+    from es_pytorch_amd.envs import make_batched
+    from es_pytorch_amd.nn.nn import FeedForward
+    from es_pytorch_amd.nn.optimizers import Adam
+    from es_pytorch_amd.parallel.comm import Comm
+    from es_pytorch_amd.utils.rankers import CenteredRanker
+
+    fits = {}
+    for mode in ("step", "episode"):
+        torch.manual_seed(70)
+        comm = Comm(dev)
+        cfg = AttrDict({"env": {"name": "Humanoid-v2", "max_steps": 30},
+                        "noise": {"tbl_size": 1_000_000, "std": 0.02},
+                        "policy": {"layer_sizes": [64], "ac_std": 0.01, "l2coeff": 0.005,
+                                   "lr": 0.01, "ob_clip": 5, "save_obs_chance": 1.0},
+                        "general": {"policies_per_gen": 8, "batch_size": 500, "seed": 2}})
+        env = make_batched("Humanoid-v2", 9, dev, max_steps=30, terminate_on_fall=False)
+        nn = FeedForward([64], torch.nn.Tanh(), env, 0.01, 5)
+        policy = Policy(nn, 0.02, Adam(len(Policy.get_flat(nn)), 0.01))
+        nt = NoiseTable.create_shared(comm, 1_000_000, len(policy), seed=3, device=dev)
+        rs = np.random.RandomState(71)
+        eng = GpuEngine(cfg, comm, policy, nt, env, rs, use_graph=True,
+                        rollout_mode=mode)
+        ranker = CenteredRanker()
+        eng.step(ranker)
+        fits[mode] = np.concatenate([ranker.fits_pos, ranker.fits_neg]).ravel()
+    np.testing.assert_allclose(fits["step"], fits["episode"], rtol=1e-5, atol=1e-5)
