@@ -29,11 +29,57 @@ _cpu_lib: Optional[ctypes.CDLL] = None
 _hip_lib: Optional[ctypes.CDLL] = None
 
 
+class _BuildLock:
+    """Serialize .so builds across processes (N torchrun ranks share the
+    tree; concurrent compiler invocations writing one output would corrupt
+    it). Uses filelock when available, else a best-effort O_EXCL spinlock."""
+
+    def __init__(self, path: str):
+        self.path = path + ".lock"
+        self._fl = None
+
+    def __enter__(self):
+        try:
+            from filelock import FileLock
+            self._fl = FileLock(self.path)
+            self._fl.acquire(timeout=600)
+        except ImportError:
+            import time
+            for _ in range(6000):
+                try:
+                    fd = os.open(self.path, os.O_CREAT | os.O_EXCL | os.O_WRONLY)
+                    os.close(fd)
+                    self._fl = "posix"
+                    break
+                except FileExistsError:
+                    time.sleep(0.1)
+        return self
+
+    def __exit__(self, *a):
+        if self._fl == "posix":
+            try:
+                os.unlink(self.path)
+            except OSError:
+                pass
+        elif self._fl is not None:
+            self._fl.release()
+
+
 def build_cpu(force: bool = False) -> str:
     src = os.path.join(_DIR, "csrc", "cpu_ops.cpp")
-    if force or not os.path.exists(_CPU_SO) or os.path.getmtime(_CPU_SO) < os.path.getmtime(src):
-        cmd = ["g++", "-O3", "-std=c++17", "-shared", "-fPIC", "-pthread", src, "-o", _CPU_SO]
-        subprocess.run(cmd, check=True, capture_output=True, text=True)
+
+    def stale():
+        return (force or not os.path.exists(_CPU_SO)
+                or os.path.getmtime(_CPU_SO) < os.path.getmtime(src))
+
+    if stale():
+        with _BuildLock(_CPU_SO):
+            if stale():  # re-check under the lock: another rank may have built
+                tmp = _CPU_SO + ".tmp"
+                cmd = ["g++", "-O3", "-std=c++17", "-shared", "-fPIC", "-pthread", src,
+                       "-o", tmp]
+                subprocess.run(cmd, check=True, capture_output=True, text=True)
+                os.replace(tmp, _CPU_SO)
     return _CPU_SO
 
 
@@ -44,16 +90,25 @@ def hip_sources():
 
 def build_hip(force: bool = False, arch: str = "gfx950") -> str:
     srcs = hip_sources()
-    newest = max(os.path.getmtime(s) for s in srcs) if srcs else 0
-    hdr = os.path.join(_DIR, "csrc", "philox.h")
-    newest = max(newest, os.path.getmtime(hdr))
-    if force or not os.path.exists(_HIP_SO) or os.path.getmtime(_HIP_SO) < newest:
-        hipcc = os.environ.get("HIPCC", "/opt/rocm/bin/hipcc")
-        cmd = [hipcc, f"--offload-arch={arch}", "-O3", "-std=c++17", "-shared", "-fPIC",
-               *srcs, "-o", _HIP_SO]
-        r = subprocess.run(cmd, check=False, capture_output=True, text=True)
-        if r.returncode != 0:
-            raise RuntimeError(f"hipcc build failed:\n{r.stdout}\n{r.stderr}")
+    hdrs = [os.path.join(_DIR, "csrc", "philox.h")] + \
+        [os.path.join(_DIR, "csrc", "hip", h)
+         for h in os.listdir(os.path.join(_DIR, "csrc", "hip")) if h.endswith(".h")]
+    newest = max(os.path.getmtime(f) for f in srcs + hdrs)
+
+    def stale():
+        return force or not os.path.exists(_HIP_SO) or os.path.getmtime(_HIP_SO) < newest
+
+    if stale():
+        with _BuildLock(_HIP_SO):
+            if stale():  # re-check under the lock: another rank may have built
+                hipcc = os.environ.get("HIPCC", "/opt/rocm/bin/hipcc")
+                tmp = _HIP_SO + ".tmp"
+                cmd = [hipcc, f"--offload-arch={arch}", "-O3", "-std=c++17", "-shared",
+                       "-fPIC", *srcs, "-o", tmp]
+                r = subprocess.run(cmd, check=False, capture_output=True, text=True)
+                if r.returncode != 0:
+                    raise RuntimeError(f"hipcc build failed:\n{r.stdout}\n{r.stderr}")
+                os.replace(tmp, _HIP_SO)
     return _HIP_SO
 
 
