@@ -129,6 +129,10 @@ def main():
         torch.cuda.synchronize(device)
     elapsed = time.perf_counter() - t0
 
+    # per-rank diagnostics to stderr (the driver parses stdout JSON only)
+    print(f"[bench rank {comm.rank}/{world}] elapsed={elapsed:.3f}s "
+          f"timings={getattr(engine, 'timings', {})}", file=sys.stderr, flush=True)
+
     # MAX elapsed over ranks = whole-job wall time
     elapsed = comm.allreduce_scalar(elapsed) if world == 1 else max(
         comm.allgather_obj(elapsed))
