@@ -42,8 +42,8 @@ def test_bench_cpu_single():
 def test_bench_torchrun_world2_cpu():
     """The exact launcher shape the driver uses, world_size=2 on CPU/gloo."""
     cmd = [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
-           "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
-           "--master-port", "29517", os.path.join(ROOT, "bench.py"), "--cpu",
+           "--nproc-per-node", "2", "--standalone", "--local-addr", "127.0.0.1",
+           os.path.join(ROOT, "bench.py"), "--cpu",
            "--gpus", "2", "--steps", "2", "--warmup", "1", "--pop-per-gpu", "4",
            "--max-steps", "20", "--env", "Hopper-v3", "--layers", "8",
            "--tbl-size", "200000"]

@@ -123,8 +123,7 @@ def test_simple_example_torchrun_world2(tmp_path):
     cfg_path = tmp_path / "cfg.json"
     cfg_path.write_text(json.dumps(cfg))
     cmd = [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
-           "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
-           "--master-port", "29523",
+           "--nproc-per-node", "2", "--standalone", "--local-addr", "127.0.0.1",
            os.path.join(ROOT, "examples", "simple_example.py"), str(cfg_path)]
     r = subprocess.run(cmd, capture_output=True, text=True, timeout=300,
                        cwd=str(tmp_path), env=dict(os.environ, PYTHONPATH=ROOT))
