@@ -44,7 +44,7 @@ from es_pytorch_amd.nn.optimizers import SGD, Adam, Optimizer
 from es_pytorch_amd.parallel.comm import Comm
 from es_pytorch_amd.rollout.results import RewardResult
 from es_pytorch_amd.utils.novelty import novelty_batch
-from es_pytorch_amd.utils.rankers import Ranker
+from es_pytorch_amd.utils.rankers import CenteredRanker, Ranker
 
 
 def forward_perm(dims: List[int]) -> torch.Tensor:
@@ -429,28 +429,56 @@ class GpuEngine:
             with torch.cuda.stream(self._side):
                 self._upload_offsets()
             self._offs_prefetched = True
-        all_rows = self.comm.allgather_rows(rows).cpu().numpy()
+        all_rows_dev = self.comm.allgather_rows(rows)
         # evaluated-episode steps only (the reference's es.py:79 counts the
         # pos+neg rollouts, not the noiseless eval)
         local_steps = float(self.member_steps[:2 * self.pairs * self.eps].sum().item())
         steps = int(self.comm.allreduce_scalar(local_steps))
         t3 = time.perf_counter()
 
-        # -- identical redundant ranking on every rank (host; pop-sized)
-        pos, neg = all_rows[:, :O], all_rows[:, O:2 * O]
-        inds = all_rows[:, -1]
-        ranker.rank(pos, neg, inds)
-
-        # -- gradient + update on device (HIP gather-GEMV + fused Adam/SGD)
-        rf = torch.from_numpy(np.ascontiguousarray(ranker.ranked_fits,
-                                                   dtype=np.float32)).to(self.device)
-        ri = torch.from_numpy(np.ascontiguousarray(ranker.noise_inds,
-                                                   dtype=np.int64)).to(self.device)
-        ops.check(ops.hip().es_grad_gather(
-            self.grad.data_ptr(), self.nt.noise.data_ptr(), rf.data_ptr(), ri.data_ptr(),
-            rf.numel(), self.n, self._stream()), "es_grad_gather")
-        self._optim_step(float(ranker.n_fits_ranked), float(cfg.policy.l2coeff))
+        # -- identical redundant ranking on every rank. K4 device path: a
+        # plain single-objective CenteredRanker runs as torch argsort on
+        # device (stable sort: deterministic, rank-identical), so the
+        # gradient + Adam launch immediately and the host mirror copy for
+        # reporters/heuristics overlaps them. Other rankers take the host
+        # path (pop-sized, sub-ms).
+        dev_rank = (type(ranker) is CenteredRanker and O == 1
+                    and self.device.type == "cuda"
+                    and all_rows_dev.device.type == "cuda")
+        if dev_rank:
+            fits_all = torch.cat([all_rows_dev[:, 0], all_rows_dev[:, 1]]).float()
+            npop = fits_all.numel()
+            r = torch.empty(npop, dtype=torch.float32, device=self.device)
+            r[torch.argsort(fits_all, stable=True)] = torch.arange(
+                npop, dtype=torch.float32, device=self.device)
+            y = r / (npop - 1) - 0.5
+            rf = (y[:npop // 2] - y[npop // 2:]).contiguous()
+            ri = all_rows_dev[:, 2].long().contiguous()
+            ops.check(ops.hip().es_grad_gather(
+                self.grad.data_ptr(), self.nt.noise.data_ptr(), rf.data_ptr(),
+                ri.data_ptr(), rf.numel(), self.n, self._stream()), "es_grad_gather")
+            self._optim_step(float(npop), float(cfg.policy.l2coeff))
+            # host mirror (for reporters and the entry scripts' heuristics)
+            # overlaps the update kernels above
+            all_rows = all_rows_dev.cpu().numpy()
+            ranker._pre_rank(all_rows[:, :O], all_rows[:, O:2 * O], all_rows[:, -1])
+            ranker.ranked_fits = rf.cpu().numpy().astype(np.float64)
+            ranker.n_fits_ranked = npop
+        else:
+            all_rows = all_rows_dev.cpu().numpy()
+            pos, neg = all_rows[:, :O], all_rows[:, O:2 * O]
+            inds = all_rows[:, -1]
+            ranker.rank(pos, neg, inds)
+            rf = torch.from_numpy(np.ascontiguousarray(ranker.ranked_fits,
+                                                       dtype=np.float32)).to(self.device)
+            ri = torch.from_numpy(np.ascontiguousarray(ranker.noise_inds,
+                                                       dtype=np.int64)).to(self.device)
+            ops.check(ops.hip().es_grad_gather(
+                self.grad.data_ptr(), self.nt.noise.data_ptr(), rf.data_ptr(),
+                ri.data_ptr(), rf.numel(), self.n, self._stream()), "es_grad_gather")
+            self._optim_step(float(ranker.n_fits_ranked), float(cfg.policy.l2coeff))
         self.sync_host(light=True)
+        pos, neg = all_rows[:, :O], all_rows[:, O:2 * O]
         t4 = time.perf_counter()
 
         # -- per-gen obstat -> merged across ranks (packed all_reduce)

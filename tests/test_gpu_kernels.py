@@ -717,3 +717,45 @@ def test_episode_rollout_mode_parity(dev):
         eng.step(ranker)
         fits[mode] = np.concatenate([ranker.fits_pos, ranker.fits_neg]).ravel()
     np.testing.assert_allclose(fits["step"], fits["episode"], rtol=1e-5, atol=1e-5)
+
+
+def test_device_rank_matches_host_rank(dev):
+    """K4 device argsort ranking == host numpy CenteredRanker (tie-free fits)."""
+    import numpy as np
+    from es_pytorch_amd.config import AttrDict
+    from es_pytorch_amd.core.engine import GpuEngine
+    from es_pytorch_amd.core.noisetable import NoiseTable
+    from es_pytorch_amd.core.policy import Policy
+    from es_pytorch_amd.envs import make_batched
+    from es_pytorch_amd.nn.nn import FeedForward
+    from es_pytorch_amd.nn.optimizers import Adam
+    from es_pytorch_amd.parallel.comm import Comm
+    from es_pytorch_amd.utils.rankers import CenteredRanker
+
+    class _HostCentered(CenteredRanker):
+        pass  # subclass -> engine's `type(...) is CenteredRanker` gate is False
+
+    thetas = {}
+    for cls in (CenteredRanker, _HostCentered):
+        torch.manual_seed(80)
+        comm = Comm(dev)
+        cfg = AttrDict({"env": {"name": "Humanoid-v2", "max_steps": 25},
+                        "noise": {"tbl_size": 1_000_000, "std": 0.02},
+                        "policy": {"layer_sizes": [64], "ac_std": 0.01, "l2coeff": 0.005,
+                                   "lr": 0.01, "ob_clip": 5, "save_obs_chance": 0.5},
+                        "general": {"policies_per_gen": 16, "batch_size": 500,
+                                    "seed": 4}})
+        env = make_batched("Humanoid-v2", 17, dev, max_steps=25,
+                           terminate_on_fall=False)
+        nn = FeedForward([64], torch.nn.Tanh(), env, 0.01, 5)
+        policy = Policy(nn, 0.02, Adam(len(Policy.get_flat(nn)), 0.01))
+        nt = NoiseTable.create_shared(comm, 1_000_000, len(policy), seed=12, device=dev)
+        rs = np.random.RandomState(81)
+        eng = GpuEngine(cfg, comm, policy, nt, env, rs, use_graph=False)
+        ranker = cls()
+        for _ in range(2):
+            eng.step(ranker)
+        thetas[cls.__name__] = policy.flat_params.copy()
+        assert eng is not None
+    np.testing.assert_allclose(thetas["CenteredRanker"], thetas["_HostCentered"],
+                               rtol=1e-6, atol=1e-7)
