@@ -129,3 +129,16 @@ def test_simple_example_torchrun_world2(tmp_path):
                        cwd=str(tmp_path), env=dict(os.environ, PYTHONPATH=ROOT))
     assert r.returncode == 0, r.stderr[-3000:]
     assert "avg fitness" in r.stdout
+
+
+def test_obj_resume_from_checkpoint(tmp_path):
+    """cfg.policy.load resume path (reference obj.py:39-41)."""
+    _run("obj.py", _base(gens=2), tmp_path)
+    weights = tmp_path / "saved" / "CartPole-v1-t" / "weights"
+    files = sorted(f for f in os.listdir(weights) if f.startswith("policy-"))
+    assert files
+    cfg = _base(gens=1)
+    cfg["policy"]["load"] = str(weights / files[-1])
+    cfg["general"]["name"] = "resumed"
+    _run("obj.py", cfg, tmp_path)
+    assert os.path.exists(tmp_path / "saved" / "CartPole-v1-resumed")
