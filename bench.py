@@ -56,6 +56,10 @@ def main():
     from es_pytorch_amd.utils.rankers import CenteredRanker
 
     use_cuda = torch.cuda.is_available() and not args.cpu
+    if not args.cpu and not use_cuda:
+        # never silently emit CPU-stand-in numbers as if they were measured
+        raise SystemExit("bench.py: no GPU visible — pass --cpu explicitly for "
+                         "the debug stand-in path")
     if args.objective == "nsr" and not use_cuda:
         raise SystemExit("--objective nsr requires the GPU engine")
     comm = init_comm()
