@@ -146,9 +146,8 @@ def main():
 
     if comm.rank == 0:
         out = {
-            "metric": ("env-steps/sec (whole node), HumanoidFlagrun NSR-A"
-                       if args.objective == "nsr" else
-                       "env-steps/sec (whole node), Humanoid-v2 ES"),
+            "metric": (f"env-steps/sec (whole node), {args.env} "
+                       f"{'NSR-A' if args.objective == 'nsr' else 'objective-ES'}"),
             "value": round(env_steps_per_sec, 1),
             "unit": "env-steps/s",
             "n_gpus": args.gpus,
