@@ -21,7 +21,7 @@ and [pairs, 2*pairs) carry -noise; slot 2*pairs is the all-noiseless game.
 from __future__ import annotations
 
 import time
-from typing import List, Optional, Tuple
+from typing import List, Tuple
 
 import numpy as np
 import torch
