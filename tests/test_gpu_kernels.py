@@ -819,3 +819,47 @@ def test_engine_sgd_optimizer(dev):
     eng.step(CenteredRanker())
     assert not np.array_equal(policy.flat_params, flat0)
     assert np.isfinite(policy.flat_params).all()
+
+
+def test_engine_all_objectives(dev):
+    """Every engine objective produces finite, sensible fitnesses."""
+    import numpy as np
+    from es_pytorch_amd.config import AttrDict
+    from es_pytorch_amd.core.engine import GpuEngine
+    from es_pytorch_amd.core.noisetable import NoiseTable
+    from es_pytorch_amd.core.policy import Policy
+    from es_pytorch_amd.envs import make_batched
+    from es_pytorch_amd.nn.nn import FeedForward
+    from es_pytorch_amd.nn.optimizers import Adam
+    from es_pytorch_amd.parallel.comm import Comm
+    from es_pytorch_amd.utils.rankers import CenteredRanker, MultiObjectiveRanker
+
+    for objective in ("reward", "mean_reward", "dist", "xdist", "ns", "nsr"):
+        torch.manual_seed(95)
+        comm = Comm(dev)
+        cfg = AttrDict({"env": {"name": "Hopper-v3", "max_steps": 15},
+                        "noise": {"tbl_size": 300_000, "std": 0.05},
+                        "policy": {"layer_sizes": [16], "ac_std": 0.0, "l2coeff": 0.005,
+                                   "lr": 0.01, "ob_clip": 5, "save_obs_chance": 1.0},
+                        "general": {"policies_per_gen": 8, "batch_size": 100,
+                                    "seed": 1}})
+        env = make_batched("Hopper-v3", 9, dev, max_steps=15, terminate_on_fall=False)
+        nn = FeedForward([16], torch.nn.Tanh(), env, 0.0, 5)
+        policy = Policy(nn, 0.05, Adam(len(Policy.get_flat(nn)), 0.01))
+        nt = NoiseTable.create_shared(comm, 300_000, len(policy), seed=14, device=dev)
+        rs = np.random.RandomState(96)
+        eng = GpuEngine(cfg, comm, policy, nt, env, rs, objective=objective,
+                        use_graph=False)
+        if objective in ("ns", "nsr"):
+            eng.archive = torch.randn(8, 2, dtype=torch.float64, device=dev)
+        ranker = MultiObjectiveRanker(CenteredRanker(), 0.5) if objective == "nsr" \
+            else CenteredRanker()
+        eng.step(ranker)
+        fits = np.concatenate([ranker.fits_pos, ranker.fits_neg])
+        assert np.isfinite(fits).all(), objective
+        if objective == "dist":
+            assert (fits >= 0).all()  # distances are non-negative
+        if objective == "ns":
+            assert (fits >= 0).all()  # novelty is a mean distance
+        if objective == "nsr":
+            assert fits.shape[1] == 2
