@@ -77,7 +77,8 @@ class GpuEngine:
     def __init__(self, cfg, comm: Comm, policy: Policy, nt: NoiseTable, env: BatchedEnv,
                  rs: np.random.RandomState, objective: str = "reward",
                  use_graph: bool = True, novelty_k: int = 10,
-                 fused: Optional[bool] = None, rollout_mode: str = "step"):
+                 fused: Optional[bool] = None, rollout_mode: str = "step",
+                 split_dyn: Optional[bool] = None):
         # rollout_mode: "step" = one kernel per env step for the population
         # (graph-replayed); "episode" = ONE kernel per generation, each block
         # runs its member's whole episode (members are mutually independent,
@@ -188,6 +189,17 @@ class GpuEngine:
             self.mo_sum = torch.zeros((self.B, D), dtype=torch.float32, device=d)
             self.mo_sumsq = torch.zeros((self.B, D), dtype=torch.float32, device=d)
 
+        # split-dynamics rollout: forward kernel + shared-A dynamics kernel
+        # per step (G members per dynamics block re-use each A octet load);
+        # bitwise-identical trajectories to the fused single-kernel step
+        self.dyn_group = int(cfg.general.get("dyn_group", 5) or 5)
+        want_split = bool(cfg.general.get("split_dyn", False)) if split_dyn is None \
+            else bool(split_dyn)
+        self.split_dyn = (want_split and self.fused and env.sdim % 8 == 0
+                          and env.ac_dim <= 64 and self.dyn_group in (2, 4, 5, 8))
+        if self.split_dyn:
+            self.act_scratch = torch.empty(self.B * 64, dtype=torch.float32, device=d)
+
     # ------------------------------------------------------------------ ops
     def _stream(self):
         return torch.cuda.current_stream(self.device).cuda_stream if \
@@ -219,7 +231,7 @@ class GpuEngine:
         main grid stays an exact multiple of the CU slot count."""
         env = self.env
         goal_ptr = env.goal.data_ptr() if env.goal_conditioned else None
-        ops.check(ops.hip().es_loco_step(
+        common = (
             self.weights.data_ptr(), self.obmean.data_ptr(), self.obstd.data_ptr(),
             self.dims_arr.ctypes.data, len(self.dims_arr), self.seed_dev.data_ptr(),
             t + 1, float(self.policy._module.ob_clip), self.acstd_dev.data_ptr(),
@@ -234,7 +246,13 @@ class GpuEngine:
             int(env.terminate_on_fall), (self.M - 1) * self.eps, self.bins, self.eps,
             self.act_mode,
             float(env.leak), float(env.ctrl_cost), float(env.alive_bonus),
-            float(env.fall_threshold), float(env.dt), self._stream()), "es_loco_step")
+            float(env.fall_threshold), float(env.dt))
+        if self.split_dyn:
+            ops.check(ops.hip().es_loco_step_split(
+                *common, self.act_scratch.data_ptr(), self.dyn_group,
+                self._stream()), "es_loco_step_split")
+        else:
+            ops.check(ops.hip().es_loco_step(*common, self._stream()), "es_loco_step")
 
     def _loco_episode(self, member_base: int, n_members: int, noiseless_from: int,
                       n_steps: Optional[int] = None, salt_base: int = 0):

@@ -43,13 +43,13 @@ struct LocoPtrs {
   float *alive, *rew_total, *member_steps, *behv, *mo_sum, *mo_sumsq;
 };
 
-__device__ __forceinline__ void loco_step_body(
+__device__ __forceinline__ void loco_fwd_body(
     const MlpShape& sh, const LocoArgs& la, const LocoPtrs& P, int b, uint64_t salt,
-    float* bufA, float* bufB, float* partial, float* raws, float* abuf, float* sc) {
+    float* bufA, float* bufB, float* partial, float* raws, float* abuf) {
   const int tid = threadIdx.x;
   const int nth = blockDim.x;
   const int S = la.S, A = la.A;
-  float* sb = P.s_glob + (int64_t)b * S;
+  const float* sb = P.s_glob + (int64_t)b * S;
 
   // ---- build normalized obs; keep raw state in LDS -----------------------
   for (int i = tid; i < S; i += nth) {
@@ -96,56 +96,69 @@ __device__ __forceinline__ void loco_step_body(
     abuf[tid] = fclampf(a, -1.0f, 1.0f);  // env action clamp (locomotion.py)
   }
   __syncthreads();
+}
 
-  // ---- dynamics + fused epilogue ----------------------------------------
-  // The epilogue that materializes s' also accumulates the reward/behaviour
-  // reduction partials AND the per-member obs statistics in the same pass
-  // (one s'-sweep instead of three, and wave shuffles replace the 8-barrier
-  // LDS reduction tree).
+// A-matvec partial sums for ONE member (bf16 A, octet-tiled like the policy
+// layers in mlp_core.h; S % 8 == 0 required). Writes the (PART, OCT, 8)
+// partial layout consumed by loco_dyn_finish's output sweep.
+__device__ __forceinline__ void loco_dyn_partials(
+    const uint16_t* Am, int S, const float* raws, float* partial, int tid, int nth) {
+  const int OCT = S >> 3;
+  const int PART = nth / OCT;
+  const int oi = tid % OCT, ip = tid / OCT;
+  float acc[8];
+#pragma unroll
+  for (int q = 0; q < 8; ++q) acc[q] = 0.0f;
+  if (ip < PART) {
+    const uint16_t* acol = Am + (oi << 3);
+    auto ld = [&](int i) {
+      return *reinterpret_cast<const uint4*>(acol + (int64_t)i * S);
+    };
+    int i = ip;
+    const int step4 = PART * 4;
+    if (i + 3 * PART < S) {
+      uint4 c0 = ld(i), c1 = ld(i + PART), c2 = ld(i + 2 * PART), c3 = ld(i + 3 * PART);
+      for (; i + 7 * PART < S; i += step4) {
+        const uint4 n0 = ld(i + 4 * PART), n1 = ld(i + 5 * PART),
+                    n2 = ld(i + 6 * PART), n3 = ld(i + 7 * PART);
+        bf8_fma(c0, raws[i], acc);
+        bf8_fma(c1, raws[i + PART], acc);
+        bf8_fma(c2, raws[i + 2 * PART], acc);
+        bf8_fma(c3, raws[i + 3 * PART], acc);
+        c0 = n0; c1 = n1; c2 = n2; c3 = n3;
+      }
+      bf8_fma(c0, raws[i], acc);
+      bf8_fma(c1, raws[i + PART], acc);
+      bf8_fma(c2, raws[i + 2 * PART], acc);
+      bf8_fma(c3, raws[i + 3 * PART], acc);
+      i += step4;
+    }
+    for (; i < S; i += PART) bf8_fma(ld(i), raws[i], acc);
+#pragma unroll
+    for (int q = 0; q < 8; ++q) partial[(ip * OCT + oi) * 8 + q] = acc[q];
+  }
+  __syncthreads();
+}
+
+// Dynamics output sweep + fused epilogue for ONE member. The sweep that
+// materializes s' also accumulates the reward/behaviour reduction partials
+// AND the per-member obs statistics in the same pass (one s'-sweep instead
+// of three, and wave shuffles replace the 8-barrier LDS reduction tree).
+// When S % 8 == 0 the A-matvec partials must already be in `partial`
+// (loco_dyn_partials); otherwise the scalar fallback reads Am directly.
+__device__ __forceinline__ void loco_dyn_finish(
+    const LocoArgs& la, const LocoPtrs& P, int b, const float* raws,
+    const float* abuf, float* partial, int tid, int nth) {
+  const int S = la.S, A = la.A;
+  float* sb = P.s_glob + (int64_t)b * S;
   const float w_alive = P.alive[b];  // pre-step alive, used as obstat weight
   float p0 = 0, p1 = 0, p2 = 0, p3 = 0;
   float* ms = P.mo_sum + (int64_t)b * la.D;
   float* mq = P.mo_sumsq + (int64_t)b * la.D;
   {
-    // bf16 A, octet-tiled like the policy layers (mlp_core.h scheme)
     const bool oct8 = (S % 8 == 0);
     const int OCT = S >> 3;
     const int PART = oct8 ? nth / OCT : 0;
-    if (oct8) {
-      const int oi = tid % OCT, ip = tid / OCT;
-      float acc[8];
-#pragma unroll
-      for (int q = 0; q < 8; ++q) acc[q] = 0.0f;
-      if (ip < PART) {
-        const uint16_t* acol = P.Am + (oi << 3);
-        auto ld = [&](int i) {
-          return *reinterpret_cast<const uint4*>(acol + (int64_t)i * S);
-        };
-        int i = ip;
-        const int step4 = PART * 4;
-        if (i + 3 * PART < S) {
-          uint4 c0 = ld(i), c1 = ld(i + PART), c2 = ld(i + 2 * PART), c3 = ld(i + 3 * PART);
-          for (; i + 7 * PART < S; i += step4) {
-            const uint4 n0 = ld(i + 4 * PART), n1 = ld(i + 5 * PART),
-                        n2 = ld(i + 6 * PART), n3 = ld(i + 7 * PART);
-            bf8_fma(c0, raws[i], acc);
-            bf8_fma(c1, raws[i + PART], acc);
-            bf8_fma(c2, raws[i + 2 * PART], acc);
-            bf8_fma(c3, raws[i + 3 * PART], acc);
-            c0 = n0; c1 = n1; c2 = n2; c3 = n3;
-          }
-          bf8_fma(c0, raws[i], acc);
-          bf8_fma(c1, raws[i + PART], acc);
-          bf8_fma(c2, raws[i + 2 * PART], acc);
-          bf8_fma(c3, raws[i + 3 * PART], acc);
-          i += step4;
-        }
-        for (; i < S; i += PART) bf8_fma(ld(i), raws[i], acc);
-#pragma unroll
-        for (int q = 0; q < 8; ++q) partial[(ip * OCT + oi) * 8 + q] = acc[q];
-      }
-      __syncthreads();
-    }
     for (int o = tid; o < S; o += nth) {
       float p = P.b0[o];
       if (oct8) {
@@ -235,6 +248,15 @@ __device__ __forceinline__ void loco_step_body(
   __syncthreads();  // LDS reuse safety for the episode kernel's next step
 }
 
+__device__ __forceinline__ void loco_step_body(
+    const MlpShape& sh, const LocoArgs& la, const LocoPtrs& P, int b, uint64_t salt,
+    float* bufA, float* bufB, float* partial, float* raws, float* abuf, float* sc) {
+  loco_fwd_body(sh, la, P, b, salt, bufA, bufB, partial, raws, abuf);
+  if (la.S % 8 == 0)
+    loco_dyn_partials(P.Am, la.S, raws, partial, threadIdx.x, blockDim.x);
+  loco_dyn_finish(la, P, b, raws, abuf, partial, threadIdx.x, blockDim.x);
+}
+
 #define ES_LOCO_CARVE()                                          \
   extern __shared__ __attribute__((aligned(16))) char smem[];    \
   float* bufA = reinterpret_cast<float*>(smem);                  \
@@ -258,6 +280,115 @@ loco_episode_kernel(MlpShape sh, LocoArgs la, LocoPtrs P, int member_base, int n
   for (int t = 1; t <= n_steps; ++t)
     loco_step_body(sh, la, P, b, (uint64_t)(salt_base + t), bufA, bufB, partial, raws,
                    abuf, sc);
+}
+
+// ---- split-dynamics path ---------------------------------------------------
+// The fused kernel re-reads the shared (S, S) bf16 transition matrix A once
+// per MEMBER (pop x 283 KB per step for Humanoid = the measured ~20 us/step
+// L2-bandwidth bound, profiles/README.md). The split path factors one env
+// step into two launches:
+//   loco_fwd_kernel  — grid = pop: obs build + policy forward + action
+//                      decode, actions written to a (pop, 64) global scratch;
+//   loco_dyn_kernel<G> — grid = pop/G: each block stages G members' states
+//                      and actions in LDS and FMAs every A octet it loads
+//                      into all G members' accumulators, cutting A traffic
+//                      G-fold. Per-member accumulation order is IDENTICAL to
+//                      the fused kernel (same (oi, ip) tiling, same depth-4
+//                      pipeline, same loco_dyn_finish), so split and fused
+//                      trajectories match bitwise.
+__global__ void __launch_bounds__(256)
+loco_fwd_kernel(MlpShape sh, LocoArgs la, LocoPtrs P, uint64_t salt, float* act_glob) {
+  ES_LOCO_CARVE();
+  (void)sc;
+  const int b = blockIdx.x;
+  loco_fwd_body(sh, la, P, b, salt, bufA, bufB, partial, raws, abuf);
+  if (threadIdx.x < la.A) act_glob[(int64_t)b * 64 + threadIdx.x] = abuf[threadIdx.x];
+}
+
+template <int G>
+__global__ void __launch_bounds__(256)
+loco_dyn_kernel(LocoArgs la, LocoPtrs P, const float* act_glob, int n_pop) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  const int S = la.S;
+  const int Spad = (S + 3) & ~3;
+  float* raws = reinterpret_cast<float*>(smem);  // G x Spad raw states
+  float* abuf = raws + G * Spad;                 // G x 64 actions
+  float* partial = abuf + G * 64;                // G x 2048 matvec partials
+  const int tid = threadIdx.x, nth = blockDim.x;
+  const int b0 = blockIdx.x * G;
+  const int gs = min(G, n_pop - b0);  // members in this block (tail block < G)
+
+  // stage states + actions; zero-fill tail slots so the accumulator loop can
+  // stay fully unrolled with compile-time member indices (register arrays)
+#pragma unroll
+  for (int g = 0; g < G; ++g) {
+    const float* sb = P.s_glob + (int64_t)(b0 + g) * S;
+    for (int i = tid; i < S; i += nth) raws[g * Spad + i] = g < gs ? sb[i] : 0.0f;
+    if (tid < la.A)
+      abuf[g * 64 + tid] = g < gs ? act_glob[(int64_t)(b0 + g) * 64 + tid] : 0.0f;
+  }
+  __syncthreads();
+
+  // shared-A sweep: one uint4 A-octet load feeds G members' bf8_fma chains
+  const int OCT = S >> 3;
+  const int PART = nth / OCT;
+  const int oi = tid % OCT, ip = tid / OCT;
+  if (ip < PART) {
+    float acc[G][8];
+#pragma unroll
+    for (int g = 0; g < G; ++g)
+#pragma unroll
+      for (int q = 0; q < 8; ++q) acc[g][q] = 0.0f;
+    const uint16_t* acol = P.Am + (oi << 3);
+    auto ld = [&](int i) {
+      return *reinterpret_cast<const uint4*>(acol + (int64_t)i * S);
+    };
+    auto fmaG = [&](const uint4& c, int i) {
+#pragma unroll
+      for (int g = 0; g < G; ++g) bf8_fma(c, raws[g * Spad + i], acc[g]);
+    };
+    int i = ip;
+    const int step4 = PART * 4;
+    if (i + 3 * PART < S) {
+      uint4 c0 = ld(i), c1 = ld(i + PART), c2 = ld(i + 2 * PART), c3 = ld(i + 3 * PART);
+      for (; i + 7 * PART < S; i += step4) {
+        const uint4 n0 = ld(i + 4 * PART), n1 = ld(i + 5 * PART),
+                    n2 = ld(i + 6 * PART), n3 = ld(i + 7 * PART);
+        fmaG(c0, i);
+        fmaG(c1, i + PART);
+        fmaG(c2, i + 2 * PART);
+        fmaG(c3, i + 3 * PART);
+        c0 = n0; c1 = n1; c2 = n2; c3 = n3;
+      }
+      fmaG(c0, i);
+      fmaG(c1, i + PART);
+      fmaG(c2, i + 2 * PART);
+      fmaG(c3, i + 3 * PART);
+      i += step4;
+    }
+    for (; i < S; i += PART) fmaG(ld(i), i);
+#pragma unroll
+    for (int g = 0; g < G; ++g)
+#pragma unroll
+      for (int q = 0; q < 8; ++q)
+        partial[g * 2048 + (ip * OCT + oi) * 8 + q] = acc[g][q];
+  }
+  __syncthreads();
+
+  for (int g = 0; g < gs; ++g)
+    loco_dyn_finish(la, P, b0 + g, raws + g * Spad, abuf + g * 64,
+                    partial + g * 2048, tid, nth);
+}
+
+template <int G>
+static int loco_launch_dyn(const LocoArgs& la, const LocoPtrs& P, const float* act_glob,
+                           int n_pop, hipStream_t stream) {
+  const int Spad = (la.S + 3) & ~3;
+  const unsigned lds = (unsigned)((G * Spad + G * 64 + G * 2048) * sizeof(float));
+  const unsigned grid = (unsigned)((n_pop + G - 1) / G);
+  loco_dyn_kernel<G><<<dim3(grid), dim3(256), lds, stream>>>(la, P, act_glob, n_pop);
+  ES_CHECK_LAUNCH();
+  return 0;
 }
 
 static int loco_prepare(MlpShape* sh, LocoArgs* la, const int32_t* dims_host, int32_t ndims,
@@ -330,6 +461,45 @@ extern "C" int es_loco_step(const void* weights, const void* obmean, const void*
       sh, la, P, salt);
   ES_CHECK_LAUNCH();
   return 0;
+}
+
+// Split-dynamics step: forward kernel (grid = pop) then shared-A dynamics
+// kernel (grid = pop/G). act_glob = (n_pop, 64) float32 scratch. Requires
+// S % 8 == 0 (octet A tiling) and G in {2, 4, 5, 8}. Bitwise-identical
+// trajectories to es_loco_step (tests/test_gpu_kernels.py).
+extern "C" int es_loco_step_split(
+    const void* weights, const void* obmean, const void* obstd,
+    const int32_t* dims_host, int32_t ndims, const void* seed_dev, uint64_t salt,
+    float ob_clip, const void* ac_std_dev, int64_t row_stride, void* s_glob, void* pos,
+    const void* goal, const void* Am, const void* Bm, const void* b0, const void* wv,
+    const void* wa, const void* wy, const void* wh, void* alive, void* rew_total,
+    void* member_steps, void* behv, void* mo_sum, void* mo_sumsq, int32_t n_pop,
+    int32_t sdim, int32_t adim, int32_t goal_flag, int32_t terminate,
+    int32_t noiseless_from, int32_t bins, int32_t eps, int32_t act_mode, float leak,
+    float ctrl, float alive_bonus, float fall_thr, float dt, void* act_glob, int32_t G,
+    void* stream) {
+  MlpShape sh;
+  LocoArgs la;
+  unsigned lds;
+  int rc = loco_prepare(&sh, &la, dims_host, ndims, row_stride, ob_clip, sdim, adim,
+                        goal_flag, terminate, noiseless_from, bins, eps, act_mode, leak,
+                        ctrl, alive_bonus, fall_thr, dt, &lds);
+  if (rc) return rc;
+  if (sdim % 8 != 0) return -106;
+  LocoPtrs P = loco_ptrs(weights, obmean, obstd, ac_std_dev, seed_dev, s_glob, pos, goal,
+                         Am, Bm, b0, wv, wa, wy, wh, alive, rew_total, member_steps, behv,
+                         mo_sum, mo_sumsq);
+  hipStream_t s = (hipStream_t)stream;
+  loco_fwd_kernel<<<dim3((unsigned)n_pop), dim3(256), lds, s>>>(sh, la, P, salt,
+                                                                (float*)act_glob);
+  ES_CHECK_LAUNCH();
+  switch (G) {
+    case 2: return loco_launch_dyn<2>(la, P, (const float*)act_glob, n_pop, s);
+    case 4: return loco_launch_dyn<4>(la, P, (const float*)act_glob, n_pop, s);
+    case 5: return loco_launch_dyn<5>(la, P, (const float*)act_glob, n_pop, s);
+    case 8: return loco_launch_dyn<8>(la, P, (const float*)act_glob, n_pop, s);
+    default: return -107;
+  }
 }
 
 extern "C" int es_loco_episode(const void* weights, const void* obmean, const void* obstd,

@@ -863,3 +863,48 @@ def test_engine_all_objectives(dev):
             assert (fits >= 0).all()  # novelty is a mean distance
         if objective == "nsr":
             assert fits.shape[1] == 2
+
+
+def test_split_dyn_bitwise_matches_fused(dev):
+    """Split-dynamics rollout (forward kernel + shared-A dynamics kernel,
+    es_loco_step_split) produces BITWISE-identical trajectories and updates
+    to the fused single-kernel step: same per-member (oi, ip) tiling, same
+    depth-4 pipeline, same loco_dyn_finish epilogue."""
+    import numpy as np
+    from es_pytorch_amd.config import AttrDict
+    from es_pytorch_amd.core.engine import GpuEngine
+    from es_pytorch_amd.core.noisetable import NoiseTable
+    from es_pytorch_amd.core.policy import Policy
+    from es_pytorch_amd.envs import make_batched
+    from es_pytorch_amd.nn.nn import FeedForward
+    from es_pytorch_amd.nn.optimizers import Adam
+    from es_pytorch_amd.parallel.comm import Comm
+    from es_pytorch_amd.utils.rankers import CenteredRanker
+
+    out = {}
+    # dyn_group=5 with 8 perturbed members -> a 3-member tail block
+    for split in (False, True):
+        torch.manual_seed(86)
+        comm = Comm(dev)
+        cfg = AttrDict({"env": {"name": "Humanoid-v2", "max_steps": 25},
+                        "noise": {"tbl_size": 1_000_000, "std": 0.02},
+                        "policy": {"layer_sizes": [64], "ac_std": 0.01, "l2coeff": 0.005,
+                                   "lr": 0.01, "ob_clip": 5, "save_obs_chance": 1.0},
+                        "general": {"policies_per_gen": 8, "batch_size": 500, "seed": 2,
+                                    "dyn_group": 5}})
+        env = make_batched("Humanoid-v2", 9, dev, max_steps=25, terminate_on_fall=True)
+        nn = FeedForward([64], torch.nn.Tanh(), env, 0.01, 5)
+        policy = Policy(nn, 0.02, Adam(len(Policy.get_flat(nn)), 0.01))
+        nt = NoiseTable.create_shared(comm, 1_000_000, len(policy), seed=3, device=dev)
+        rs = np.random.RandomState(87)
+        eng = GpuEngine(cfg, comm, policy, nt, env, rs, use_graph=True, split_dyn=split)
+        assert eng.split_dyn == split
+        ranker = CenteredRanker()
+        for _ in range(2):
+            eng.step(ranker)
+        torch.cuda.synchronize(dev)
+        out[split] = (eng.theta.cpu().clone(), eng.rew_total.cpu().clone(),
+                      np.concatenate([ranker.fits_pos, ranker.fits_neg]).ravel())
+    assert torch.equal(out[False][0], out[True][0])  # params bitwise equal
+    assert torch.equal(out[False][1], out[True][1])  # rewards bitwise equal
+    np.testing.assert_array_equal(out[False][2], out[True][2])

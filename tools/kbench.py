@@ -27,6 +27,9 @@ def main():
     p.add_argument("--graph", action="store_true")
     p.add_argument("--episode", action="store_true", help="whole-episode rollout mode")
     p.add_argument("--chunk", type=int, default=1, help="env steps per launch")
+    p.add_argument("--split", action="store_true", help="split-dynamics rollout")
+    p.add_argument("--dyn-group", type=int, default=5,
+                   help="members per dynamics block in split mode")
     args = p.parse_args()
 
     from es_pytorch_amd.config import AttrDict
@@ -48,7 +51,7 @@ def main():
         "policy": {"layer_sizes": list(args.layers), "ac_std": 0.01, "l2coeff": 0.005,
                    "lr": 0.01, "ob_clip": 5, "save_obs_chance": 0.01},
         "general": {"policies_per_gen": args.pop, "batch_size": 500, "seed": 5,
-                    "steps_per_launch": args.chunk},
+                    "steps_per_launch": args.chunk, "dyn_group": args.dyn_group},
     })
     B = args.pop + 1
     env = make_batched(args.env, B, dev, max_steps=args.steps, terminate_on_fall=False)
@@ -58,7 +61,8 @@ def main():
                                   device=dev)
     rs = np.random.RandomState(0)
     eng = GpuEngine(cfg, comm, policy, nt, env, rs, use_graph=args.graph,
-                    rollout_mode="episode" if args.episode else "step")
+                    rollout_mode="episode" if args.episode else "step",
+                    split_dyn=args.split or None)
     ranker = CenteredRanker()
     eng.step(ranker)  # warmup
 
