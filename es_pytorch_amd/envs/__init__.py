@@ -32,6 +32,9 @@ def make_batched(name: str, batch: int, device="cpu", **kwargs) -> BatchedEnv:
     if c in ("HumanoidFlagrun", "HumanoidFlagrunHarder"):
         kwargs.setdefault("goal_conditioned", True)
         return SyntheticLocomotion("HumanoidFlagrun", batch, device, **kwargs)
+    if c in ("AntFlagrun", "AntGather"):  # reference flagrun.py's AntGather env
+        kwargs.setdefault("goal_conditioned", True)
+        return SyntheticLocomotion("AntFlagrun", batch, device, **kwargs)
     if c in LOCO_SHAPES:
         return SyntheticLocomotion(c, batch, device, **kwargs)
     raise ValueError(f"unknown env: {name!r} (canonical {c!r}); "

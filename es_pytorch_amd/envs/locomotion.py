@@ -36,8 +36,13 @@ LOCO_SHAPES = {
     "Walker2d": (17, 6),
     "HalfCheetah": (17, 6),
     "Ant": (27, 8),
+    "Swimmer": (8, 2),
+    "Reacher": (11, 2),
+    "InvertedPendulum": (4, 1),
+    "InvertedDoublePendulum": (11, 1),
     "Humanoid": (376, 17),
     "HumanoidFlagrun": (378, 17),  # Humanoid + 2 goal-relative dims
+    "AntFlagrun": (29, 8),         # Ant + 2 goal-relative dims (AntGather-style)
 }
 
 

@@ -9,9 +9,21 @@ from es_pytorch_amd.envs.locomotion import LOCO_SHAPES, SyntheticLocomotion
 
 def test_registry_aliases():
     for name in ["CartPole-v1", "HopperBulletEnv-v0", "Hopper-v3", "Humanoid-v2",
-                 "HumanoidFlagrunBulletEnv-v0"]:
+                 "HumanoidFlagrunBulletEnv-v0", "AntGatherBulletEnv-v0",
+                 "Swimmer-v3", "Reacher-v2", "InvertedDoublePendulum-v2"]:
         env = make(name)
         assert env.observation_space.shape[0] > 0
+
+
+def test_every_registered_env_steps():
+    import torch
+    from es_pytorch_amd.envs import make_batched, registry
+    for name in registry():
+        env = make_batched(name, 3, "cpu")
+        ob = env.reset(0)
+        assert ob.shape == (3, env.ob_dim), name
+        ob2, rew, done = env.step(torch.zeros((3, env.ac_dim)))
+        assert ob2.shape == (3, env.ob_dim) and rew.shape == (3,), name
 
 
 def test_cartpole_api_and_termination():
