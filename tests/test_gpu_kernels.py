@@ -908,3 +908,14 @@ def test_split_dyn_bitwise_matches_fused(dev):
     assert torch.equal(out[False][0], out[True][0])  # params bitwise equal
     assert torch.equal(out[False][1], out[True][1])  # rewards bitwise equal
     np.testing.assert_array_equal(out[False][2], out[True][2])
+
+
+def test_new_env_shapes_parity(dev):
+    """Swimmer (S=8: OCT=1 octet dynamics tiling), InvertedPendulum (S=4:
+    scalar fallback), AntFlagrun (goal-conditioned, S=27): fused == torch."""
+    import numpy as np
+    for env_name, layers in (("Swimmer-v3", [32]), ("InvertedPendulum-v2", [16]),
+                             ("AntFlagrun-v3", [32])):
+        fits = _engine_pair(dev, env_name, layers, pop=6, max_steps=15, seed=61)
+        np.testing.assert_allclose(fits[False], fits[True], rtol=1e-3, atol=1e-2,
+                                   err_msg=env_name)
