@@ -613,3 +613,39 @@ class GpuEngine:
                 inv[self.perm] = self.v
                 opt.v[:] = inv.cpu().numpy()
             self.policy.set_nn_params(self.policy.flat_params)
+
+    def restore_from_policy(self, gen: Optional[int] = None):
+        """Re-push host Policy state into the device buffers after an
+        in-place checkpoint restore (utils/checkpoint.py): params, optimizer
+        moments, obs normalization, and optionally the generation counter
+        (which seeds per-gen env variations, so exact resume needs it)."""
+        flat = torch.from_numpy(self.policy.flat_params).to(self.device)
+        self.theta.copy_(flat[self.perm])
+        self.m.zero_()
+        self.v.zero_()
+        self._load_optim_state()
+        self._push_obstat()
+        self._offs_prefetched = False  # pre-restore offset draws are stale
+        if gen is not None:
+            self.gen = int(gen)
+
+    def checkpoint_state(self) -> dict:
+        """Per-rank engine internals for RunCheckpointer beyond the Policy:
+        the generation counter and the side-stream-prefetched next-gen noise
+        offsets. The prefetch consumes rs draws at the END of a step, so a
+        resume that re-drew them would desync from the uninterrupted RNG
+        stream — the drawn values must travel with the snapshot."""
+        st = {"gen": self.gen, "prefetched": None}
+        if self._offs_prefetched:
+            if self._side is not None:
+                torch.cuda.current_stream(self.device).wait_stream(self._side)
+            st["prefetched"] = self.offsets[:self.pairs].cpu().numpy()
+        return st
+
+    def load_checkpoint_state(self, st: dict):
+        self.gen = int(st["gen"])
+        if st.get("prefetched") is not None:
+            offs = torch.from_numpy(np.asarray(st["prefetched"])).to(self.device)
+            self.offsets[:self.pairs].copy_(offs)
+            self.offsets[self.pairs:2 * self.pairs].copy_(offs)
+            self._offs_prefetched = True

@@ -1,0 +1,162 @@
+"""Run-level checkpointing: an atomic ring of full-run snapshots + exact resume.
+
+Beyond reference parity: the reference checkpoints only the Policy pickle
+(``src/utils/reporters.py:177-188``, resumed via ``obj.py:39-41``), so a
+resumed run repeats neither schedules nor RNG streams. A RunCheckpointer
+snapshot carries EVERYTHING the next generation depends on — the Policy
+pickle bytes (params, ObStat, optimizer state; reference format), every
+rank's numpy RandomState and torch RNG state, the mutated config scalars
+(lr/std/ac_std decay schedules), the generation index and arbitrary extra
+state (novelty archive, stagnation counters) — so a killed run resumes
+bit-for-bit: N generations straight equals k generations + kill + resume +
+N-k generations (``tests/test_checkpoint.py``).
+
+Write protocol (rank 0): gather per-rank RNG states over the collective,
+pickle to ``ckpt-<gen>.pkl.tmp``, fsync, ``os.replace`` into place — a crash
+mid-write can never corrupt the newest good snapshot — then prune the ring
+to the ``keep`` newest. All ranks barrier after the write so no rank runs
+ahead of durable state.
+"""
+from __future__ import annotations
+
+import os
+import pickle
+import re
+from typing import Any, Dict, Optional, Tuple
+
+import numpy as np
+import torch
+
+from es_pytorch_amd.config import merge_override
+from es_pytorch_amd.core.policy import Policy
+from es_pytorch_amd.parallel.comm import Comm
+
+_CKPT_RE = re.compile(r"^ckpt-(\d+)\.pkl$")
+
+
+class RunCheckpointer:
+    def __init__(self, folder: str, comm: Comm, keep: int = 3, every: int = 1):
+        self.folder = folder
+        self.comm = comm
+        self.keep = max(1, int(keep))
+        self.every = max(1, int(every))
+
+    # ----------------------------------------------------------------- save
+    def maybe_save(self, next_gen: int, policy: Policy, rs: np.random.RandomState,
+                   cfg=None, engine=None, env=None,
+                   extra: Optional[Dict[str, Any]] = None) -> bool:
+        """Save iff ``next_gen`` lands on the ``every`` cadence."""
+        if next_gen % self.every != 0:
+            return False
+        self.save(next_gen, policy, rs, cfg=cfg, engine=engine, env=env, extra=extra)
+        return True
+
+    def save(self, next_gen: int, policy: Policy, rs: np.random.RandomState,
+             cfg=None, engine=None, env=None,
+             extra: Optional[Dict[str, Any]] = None) -> str:
+        """Snapshot state such that the run continues from generation
+        ``next_gen`` exactly as if it had never stopped."""
+        if engine is not None:
+            engine.sync_host(light=False)  # device truth -> Policy incl. moments
+        # identical updates are recomputed on every rank (reference es.py:98-101
+        # design), so rank 0's policy is THE policy; RNG streams are per-rank
+        rng_states = self.comm.allgather_obj((rs.get_state(),
+                                              torch.get_rng_state().numpy()))
+        # per-rank engine internals (gen counter + prefetched offset draws)
+        eng_states = self.comm.allgather_obj(
+            engine.checkpoint_state() if engine is not None else None)
+        path = os.path.join(self.folder, f"ckpt-{next_gen}.pkl")
+        if self.comm.rank == 0:
+            os.makedirs(self.folder, exist_ok=True)
+            state = {
+                "next_gen": int(next_gen),
+                "world_size": self.comm.size,
+                "policy": pickle.dumps(policy),
+                "rng_states": rng_states,
+                "cfg": _plain(cfg) if cfg is not None else None,
+                "engine_states": eng_states,
+                # episodic CPU envs count episodes via a per-env seed counter
+                # (envs/base.py SingleFromBatched._seed); GPU envs are
+                # reseeded from engine.gen each generation and need nothing
+                "env_seed": getattr(env, "_seed", None),
+                "extra": extra or {},
+            }
+            tmp = path + ".tmp"
+            with open(tmp, "wb") as f:
+                pickle.dump(state, f)
+                f.flush()
+                os.fsync(f.fileno())
+            os.replace(tmp, path)
+            self._prune()
+        self.comm.barrier()
+        return path
+
+    def _prune(self):
+        snaps = self._snapshots()
+        for _, p in snaps[:-self.keep]:
+            try:
+                os.unlink(p)
+            except OSError:
+                pass
+
+    def _snapshots(self):
+        if not os.path.isdir(self.folder):
+            return []
+        out = []
+        for f in os.listdir(self.folder):
+            m = _CKPT_RE.match(f)
+            if m:
+                out.append((int(m.group(1)), os.path.join(self.folder, f)))
+        return sorted(out)
+
+    # -------------------------------------------------------------- restore
+    def latest(self) -> Optional[str]:
+        snaps = self._snapshots()
+        return snaps[-1][1] if snaps else None
+
+    def load(self, path: Optional[str] = None) -> Optional[Dict[str, Any]]:
+        path = path or self.latest()
+        if path is None:
+            return None
+        with open(path, "rb") as f:
+            return pickle.load(f)
+
+    def restore(self, state: Dict[str, Any], policy: Policy,
+                rs: np.random.RandomState, cfg=None, engine=None,
+                env=None) -> Tuple[int, Dict[str, Any]]:
+        """Restore IN PLACE into the live objects; returns (next_gen, extra)."""
+        if state["world_size"] != self.comm.size:
+            raise RuntimeError(
+                f"checkpoint was written by world_size={state['world_size']}, "
+                f"resuming with {self.comm.size} ranks would desync the RNG "
+                "streams — relaunch with the original rank count")
+        saved: Policy = pickle.loads(state["policy"])
+        policy.flat_params[:] = saved.flat_params
+        policy.std = saved.std
+        policy.optim.__dict__.update(saved.optim.__dict__)
+        policy.obstat = saved.obstat
+        if hasattr(saved._module, "_action_std"):
+            policy._module._action_std = saved._module._action_std
+        policy.set_nn_params(policy.flat_params)
+        np_state, torch_state = state["rng_states"][self.comm.rank]
+        rs.set_state(np_state)
+        torch.set_rng_state(torch.from_numpy(torch_state))
+        if cfg is not None and state["cfg"] is not None:
+            merge_override(cfg, state["cfg"])
+        if engine is not None:
+            engine.restore_from_policy(gen=state["next_gen"])
+            est = state.get("engine_states", [None] * self.comm.size)[self.comm.rank]
+            if est is not None:
+                engine.load_checkpoint_state(est)
+        if env is not None and state.get("env_seed") is not None:
+            env._seed = state["env_seed"]
+        return state["next_gen"], dict(state["extra"])
+
+
+def _plain(obj):
+    """AttrDict/dict tree -> plain-dict tree (stable pickles)."""
+    if isinstance(obj, dict):
+        return {k: _plain(v) for k, v in obj.items()}
+    if isinstance(obj, (list, tuple)):
+        return type(obj)(_plain(v) for v in obj)
+    return obj

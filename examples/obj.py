@@ -51,7 +51,27 @@ def main(cfg):
     time_since_best = 0
     noise_std_inc = 0.08
 
-    for gen in range(cfg.general.gens):
+    # atomic ring checkpointing + exact resume (utils/checkpoint.py); a
+    # relaunch of the same command continues bit-for-bit from the newest
+    # snapshot (beyond the reference's policy-only checkpoints, obj.py:39-41)
+    ck_every = int(cfg.general.get("ckpt_every", 0) or 0)
+    ckpt, start_gen = None, 0
+    if ck_every:
+        from es_pytorch_amd.utils.checkpoint import RunCheckpointer
+        ckpt = RunCheckpointer(path.join("saved", full_name, "ckpt"), comm,
+                               keep=int(cfg.general.get("ckpt_keep", 3)),
+                               every=ck_every)
+        state = ckpt.load()
+        if state is not None:
+            start_gen, extra = ckpt.restore(state, policy, rs, cfg=cfg,
+                                            engine=engine, env=env)
+            best_max_rew = extra.get("best_max_rew", best_max_rew)
+            time_since_best = extra.get("time_since_best", time_since_best)
+            if 0 < elite < 1:
+                ranker.elite_percent = extra.get("elite_percent", ranker.elite_percent)
+            reporter.print(f"resumed from checkpoint at gen {start_gen}")
+
+    for gen in range(start_gen, cfg.general.gens):
         if mlflow_reporter is not None:
             mlflow_reporter.set_active_run(0)
         reporter.start_gen()
@@ -114,6 +134,13 @@ def main(cfg):
             reporter.print(f"saving max policy with rew:{best_max_rew:0.2f}")
 
         reporter.end_gen()
+
+        if ckpt is not None:
+            extra = {"best_max_rew": best_max_rew, "time_since_best": time_since_best}
+            if 0 < elite < 1:
+                extra["elite_percent"] = ranker.elite_percent
+            ckpt.maybe_save(gen + 1, policy, rs, cfg=cfg, engine=engine, env=env,
+                            extra=extra)
 
     if engine is not None:
         engine.sync_host()

@@ -1,0 +1,106 @@
+"""RunCheckpointer (utils/checkpoint.py): atomic ring + bit-exact resume.
+
+The headline property: N generations straight produces bitwise-identical
+parameters to k generations + save + restore-into-fresh-objects + N-k
+generations — params, optimizer moments, ObStat, RNG streams, decay
+schedules and the env episode-seed counter all survive the round trip.
+"""
+import os
+
+import numpy as np
+import pytest
+import torch
+
+from es_pytorch_amd.config import AttrDict
+from es_pytorch_amd.core import es
+from es_pytorch_amd.core.noisetable import NoiseTable
+from es_pytorch_amd.core.policy import Policy
+from es_pytorch_amd.envs import make
+from es_pytorch_amd.nn.nn import FeedForward
+from es_pytorch_amd.nn.optimizers import Adam
+from es_pytorch_amd.parallel.comm import Comm
+from es_pytorch_amd.rollout import RewardResult, run_model
+from es_pytorch_amd.utils.checkpoint import RunCheckpointer
+from es_pytorch_amd.utils.rankers import CenteredRanker
+from es_pytorch_amd.utils.reporters import StdoutReporter
+
+
+def _fresh():
+    torch.manual_seed(3)
+    comm = Comm(torch.device("cpu"))
+    cfg = AttrDict({"general": {"policies_per_gen": 8, "batch_size": 100},
+                    "policy": {"l2coeff": 0.005},
+                    "noise": {"std": 0.05, "std_decay": 0.99, "std_limit": 0.01}})
+    env = make("CartPole-v1")
+    env.seed(0)
+    rs = np.random.RandomState(7)
+    nn = FeedForward([8], torch.nn.Tanh(), env, ac_std=0.01, ob_clip=5)
+    policy = Policy(nn, cfg.noise.std, Adam(len(Policy.get_flat(nn)), 0.05))
+    nt = NoiseTable(len(policy), NoiseTable.make_noise(100_000, seed=2))
+    return comm, cfg, env, rs, policy, nt, CenteredRanker()
+
+
+def _run(objs, n):
+    comm, cfg, env, rs, policy, nt, ranker = objs
+    reporter = StdoutReporter(comm)
+
+    def fit_fn(model, use_noise=True):
+        rews, behv, obs, steps = run_model(model, env, 80, rs if use_noise else None)
+        return RewardResult(rews, behv, obs, steps)
+
+    for _ in range(n):
+        tr, gen_obstat = es.step(cfg, comm, policy, nt, env, fit_fn, rs, ranker,
+                                 reporter)
+        policy.update_obstat(gen_obstat)
+        # schedule mutation like obj.py (must survive the resume too)
+        cfg.noise.std = policy.std = max(cfg.noise.std * cfg.noise.std_decay,
+                                         cfg.noise.std_limit)
+
+
+def test_exact_resume(tmp_path):
+    a = _fresh()
+    _run(a, 6)
+
+    b = _fresh()
+    _run(b, 3)
+    comm, cfg, env, rs, policy, nt, _ = b
+    ck = RunCheckpointer(str(tmp_path / "ring"), comm, keep=2)
+    ck.save(3, policy, rs, cfg=cfg, env=env)
+
+    c = _fresh()
+    comm2, cfg2, env2, rs2, policy2, nt2, _ = c
+    ck2 = RunCheckpointer(str(tmp_path / "ring"), comm2, keep=2)
+    state = ck2.load()
+    assert state is not None
+    next_gen, extra = ck2.restore(state, policy2, rs2, cfg=cfg2, env=env2)
+    assert next_gen == 3
+    _run(c, 3)
+
+    ap, cp = a[4], policy2
+    np.testing.assert_array_equal(ap.flat_params, cp.flat_params)
+    assert ap.std == cp.std and ap.optim.t == cp.optim.t
+    np.testing.assert_array_equal(ap.optim.m, cp.optim.m)
+    assert ap.obstat.count == cp.obstat.count
+    np.testing.assert_array_equal(np.asarray(ap.obstat.mean),
+                                  np.asarray(cp.obstat.mean))
+
+
+def test_ring_prune_and_atomicity(tmp_path):
+    comm, cfg, env, rs, policy, nt, _ = _fresh()
+    ck = RunCheckpointer(str(tmp_path / "ring"), comm, keep=2, every=2)
+    saved = [g for g in range(1, 7) if ck.maybe_save(g, policy, rs)]
+    assert saved == [2, 4, 6]  # cadence respected
+    names = sorted(os.listdir(tmp_path / "ring"))
+    assert names == ["ckpt-4.pkl", "ckpt-6.pkl"]  # ring pruned to keep=2
+    assert not any(n.endswith(".tmp") for n in names)  # atomic writes
+    assert ck.latest().endswith("ckpt-6.pkl")
+
+
+def test_world_size_mismatch_rejected(tmp_path):
+    comm, cfg, env, rs, policy, nt, _ = _fresh()
+    ck = RunCheckpointer(str(tmp_path / "ring"), comm)
+    ck.save(1, policy, rs)
+    state = ck.load()
+    state["world_size"] = 8
+    with pytest.raises(RuntimeError, match="world_size"):
+        ck.restore(state, policy, rs)

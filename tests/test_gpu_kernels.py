@@ -919,3 +919,73 @@ def test_new_env_shapes_parity(dev):
         fits = _engine_pair(dev, env_name, layers, pop=6, max_steps=15, seed=61)
         np.testing.assert_allclose(fits[False], fits[True], rtol=1e-3, atol=1e-2,
                                    err_msg=env_name)
+
+
+def test_engine_exact_checkpoint_resume(dev, tmp_path):
+    """RunCheckpointer on the GPU engine: 4 gens straight == 2 gens + save +
+    restore-into-fresh-engine + 2 gens, bitwise (params, moments, rewards).
+    Covers the prefetched-offsets hazard: the side-stream offset prefetch
+    consumes rs draws at the end of a step, and the snapshot must carry the
+    drawn values rather than let the resumed run re-draw."""
+    import numpy as np
+    from es_pytorch_amd.config import AttrDict
+    from es_pytorch_amd.core.engine import GpuEngine
+    from es_pytorch_amd.core.noisetable import NoiseTable
+    from es_pytorch_amd.core.policy import Policy
+    from es_pytorch_amd.envs import make_batched
+    from es_pytorch_amd.nn.nn import FeedForward
+    from es_pytorch_amd.nn.optimizers import Adam
+    from es_pytorch_amd.parallel.comm import Comm
+    from es_pytorch_amd.utils.checkpoint import RunCheckpointer
+    from es_pytorch_amd.utils.rankers import CenteredRanker
+
+    def fresh():
+        torch.manual_seed(50)
+        comm = Comm(dev)
+        cfg = AttrDict({"env": {"name": "Hopper-v3", "max_steps": 20},
+                        "noise": {"tbl_size": 400_000, "std": 0.02},
+                        "policy": {"layer_sizes": [32], "ac_std": 0.01,
+                                   "l2coeff": 0.005, "lr": 0.01, "ob_clip": 5,
+                                   "save_obs_chance": 0.5},
+                        "general": {"policies_per_gen": 8, "batch_size": 500,
+                                    "seed": 4}})
+        env = make_batched("Hopper-v3", 9, dev, max_steps=20)
+        nn = FeedForward([32], torch.nn.Tanh(), env, 0.01, 5)
+        policy = Policy(nn, 0.02, Adam(len(Policy.get_flat(nn)), 0.01))
+        nt = NoiseTable.create_shared(comm, 400_000, len(policy), seed=5, device=dev)
+        rs = np.random.RandomState(51)
+        eng = GpuEngine(cfg, comm, policy, nt, env, rs, use_graph=True)
+        return comm, cfg, env, rs, policy, eng
+
+    def run(objs, n):
+        *_, eng = objs
+        ranker = CenteredRanker()
+        obstats = []
+        for _ in range(n):
+            tr, ob = eng.step(ranker)
+            eng.update_obstat(ob)
+            obstats.append(tr.reward)
+        return obstats
+
+    a = fresh()
+    rews_a = run(a, 4)
+    a[5].sync_host(light=False)
+
+    b = fresh()
+    run(b, 2)
+    ck = RunCheckpointer(str(tmp_path / "ring"), b[0])
+    ck.save(2, b[4], b[3], cfg=b[1], engine=b[5])
+
+    c = fresh()
+    comm_c, cfg_c, env_c, rs_c, policy_c, eng_c = c
+    ck2 = RunCheckpointer(str(tmp_path / "ring"), comm_c)
+    state = ck2.load()
+    next_gen, _ = ck2.restore(state, policy_c, rs_c, cfg=cfg_c, engine=eng_c)
+    assert next_gen == 2 and eng_c.gen == 2
+    rews_c = run(c, 2)
+    eng_c.sync_host(light=False)
+
+    np.testing.assert_array_equal(a[4].flat_params, policy_c.flat_params)
+    np.testing.assert_array_equal(a[4].optim.m, policy_c.optim.m)
+    assert a[4].optim.t == policy_c.optim.t
+    assert rews_a[2:] == rews_c  # resumed generations bitwise-reproduce
