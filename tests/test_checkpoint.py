@@ -104,3 +104,42 @@ def test_world_size_mismatch_rejected(tmp_path):
     state["world_size"] = 8
     with pytest.raises(RuntimeError, match="world_size"):
         ck.restore(state, policy, rs)
+
+
+def _ckpt_roundtrip_rank(rank, world, folder):
+    import torch as T
+
+    from es_pytorch_amd.nn.optimizers import Adam as A2
+    from es_pytorch_amd.parallel.comm import Comm as C2
+
+    T.manual_seed(1)
+    env = make("CartPole-v1")
+    nn = FeedForward([4], T.nn.Tanh(), env, ac_std=0.0, ob_clip=5)
+    policy = Policy(nn, 0.02, A2(len(Policy.get_flat(nn)), 0.01))
+    comm = C2(T.device("cpu"))
+    rs = np.random.RandomState(100 + rank)  # distinct per-rank stream
+    rs.randint(0, 1000, size=5)
+    policy.flat_params[:] = rank + 1.0
+
+    ck = RunCheckpointer(folder, comm, keep=2)
+    ck.save(7, policy, rs)
+    expected = rs.randint(0, 10 ** 6, size=8)  # what the live run draws next
+
+    policy.flat_params[:] = -1.0  # dirty the state
+    rs2 = np.random.RandomState(0)
+    ck2 = RunCheckpointer(folder, comm, keep=2)
+    next_gen, _ = ck2.restore(ck2.load(), policy, rs2)
+    assert next_gen == 7
+    # rank 0 wrote the file; every rank restores ITS OWN rng stream slice
+    got = rs2.randint(0, 10 ** 6, size=8)
+    np.testing.assert_array_equal(got, expected)
+    # policy comes from rank 0's copy (updates are rank-identical by design)
+    np.testing.assert_array_equal(policy.flat_params,
+                                  np.full_like(policy.flat_params, 1.0))
+    return int(expected[0])
+
+
+def test_checkpoint_world2(tmp_path):
+    from tests.mp_helpers import run_mp
+    draws = run_mp(_ckpt_roundtrip_rank, world=2, args=(str(tmp_path / "ring"),))
+    assert draws[0] != draws[1]  # streams really are per-rank
