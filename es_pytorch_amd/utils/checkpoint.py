@@ -21,13 +21,13 @@ from __future__ import annotations
 
 import os
 import pickle
+import random
 import re
 from typing import Any, Dict, Optional, Tuple
 
 import numpy as np
 import torch
 
-from es_pytorch_amd.config import merge_override
 from es_pytorch_amd.core.policy import Policy
 from es_pytorch_amd.parallel.comm import Comm
 
@@ -56,15 +56,18 @@ class RunCheckpointer:
              extra: Optional[Dict[str, Any]] = None) -> str:
         """Snapshot state such that the run continues from generation
         ``next_gen`` exactly as if it had never stopped."""
-        if engine is not None:
-            engine.sync_host(light=False)  # device truth -> Policy incl. moments
+        engines = _as_list(engine)
+        for e in engines:
+            e.sync_host(light=False)  # device truth -> Policy incl. moments
         # identical updates are recomputed on every rank (reference es.py:98-101
         # design), so rank 0's policy is THE policy; RNG streams are per-rank
+        # (numpy rs, torch, and python random — nsra.py's policy selection)
         rng_states = self.comm.allgather_obj((rs.get_state(),
-                                              torch.get_rng_state().numpy()))
+                                              torch.get_rng_state().numpy(),
+                                              random.getstate()))
         # per-rank engine internals (gen counter + prefetched offset draws)
         eng_states = self.comm.allgather_obj(
-            engine.checkpoint_state() if engine is not None else None)
+            [e.checkpoint_state() for e in engines] or None)
         path = os.path.join(self.folder, f"ckpt-{next_gen}.pkl")
         if self.comm.rank == 0:
             os.makedirs(self.folder, exist_ok=True)
@@ -130,27 +133,47 @@ class RunCheckpointer:
                 f"checkpoint was written by world_size={state['world_size']}, "
                 f"resuming with {self.comm.size} ranks would desync the RNG "
                 "streams — relaunch with the original rank count")
-        saved: Policy = pickle.loads(state["policy"])
-        policy.flat_params[:] = saved.flat_params
-        policy.std = saved.std
-        policy.optim.__dict__.update(saved.optim.__dict__)
-        policy.obstat = saved.obstat
-        if hasattr(saved._module, "_action_std"):
-            policy._module._action_std = saved._module._action_std
-        policy.set_nn_params(policy.flat_params)
-        np_state, torch_state = state["rng_states"][self.comm.rank]
+        saved = pickle.loads(state["policy"])
+        for live, snap in zip(_as_list(policy), _as_list(saved)):
+            _restore_policy(live, snap)
+        np_state, torch_state, py_state = state["rng_states"][self.comm.rank]
         rs.set_state(np_state)
         torch.set_rng_state(torch.from_numpy(torch_state))
+        random.setstate(py_state)
         if cfg is not None and state["cfg"] is not None:
-            merge_override(cfg, state["cfg"])
-        if engine is not None:
-            engine.restore_from_policy(gen=state["next_gen"])
-            est = state.get("engine_states", [None] * self.comm.size)[self.comm.rank]
+            # restore ONLY the schedule-mutated scalars: run-duration and
+            # launch intent (gens, ckpt cadence, env, ...) must come from the
+            # NEW invocation, or a relaunch with a larger gens budget would
+            # be clobbered back to the finished one and exit immediately
+            for sect, key in (("noise", "std"), ("policy", "lr"),
+                              ("policy", "ac_std")):
+                sv = state["cfg"].get(sect, {})
+                if key in sv and sect in cfg and key in cfg[sect]:
+                    cfg[sect][key] = sv[key]
+        ests = state.get("engine_states", [None] * self.comm.size)[self.comm.rank]
+        for e, est in zip(_as_list(engine), ests or []):
+            e.restore_from_policy(gen=state["next_gen"])
             if est is not None:
-                engine.load_checkpoint_state(est)
+                e.load_checkpoint_state(est)
         if env is not None and state.get("env_seed") is not None:
             env._seed = state["env_seed"]
         return state["next_gen"], dict(state["extra"])
+
+
+def _as_list(x):
+    if x is None:
+        return []
+    return list(x) if isinstance(x, (list, tuple)) else [x]
+
+
+def _restore_policy(policy: Policy, saved: Policy):
+    policy.flat_params[:] = saved.flat_params
+    policy.std = saved.std
+    policy.optim.__dict__.update(saved.optim.__dict__)
+    policy.obstat = saved.obstat
+    if hasattr(saved._module, "_action_std"):
+        policy._module._action_std = saved._module._action_std
+    policy.set_nn_params(policy.flat_params)
 
 
 def _plain(obj):

@@ -98,9 +98,23 @@ def main(cfg):
     def ns_fn(model, use_ac_noise=True):
         return episodic_fit_fn(cfg, env, rs, NSRResult, archive_box)(model, use_ac_noise)
 
+    # atomic ring checkpointing + exact resume (utils/checkpoint.py): the
+    # whole NSRA control state (archive, per-policy weights/novelties,
+    # selection RNG) survives a kill (cfg.general.ckpt_every)
+    ck_every = int(cfg.general.get("ckpt_every", 0) or 0)
+    ckpt, start_gen, ck_state = None, 0, None
+    if ck_every:
+        from es_pytorch_amd.utils.checkpoint import RunCheckpointer
+        ckpt = RunCheckpointer(path.join("saved", full_name, "ckpt"), comm,
+                               keep=int(cfg.general.get("ckpt_keep", 3)),
+                               every=ck_every)
+        ck_state = ckpt.load()
+
     # ---- archive init (reference nsra.py:31-45) ----------------------------
     policies_novelties = []
-    if use_gpu:
+    if ck_state is not None:
+        pass  # resumed runs restore the archive instead of re-initializing
+    elif use_gpu:
         # one cheap generation per policy yields its noiseless behaviour
         ranker0 = MultiObjectiveRanker(CenteredRanker(), cfg.nsr.initial_w)
         for i, eng in enumerate(engines):
@@ -127,7 +141,19 @@ def main(cfg):
     obj_weight = [cfg.nsr.initial_w] * cfg.general.n_policies
     best_rew, best_dist = -np.inf, -np.inf
 
-    for gen in range(cfg.general.gens):
+    if ck_state is not None:
+        start_gen, extra = ckpt.restore(ck_state, population, rs, cfg=cfg,
+                                        engine=engines or None)
+        archive = np.asarray(extra["archive"])
+        archive_box["archive"] = archive
+        policies_novelties = extra["policies_novelties"]
+        policies_best_rewards = extra["policies_best_rewards"]
+        time_since_best = extra["time_since_best"]
+        obj_weight = extra["obj_weight"]
+        best_rew, best_dist = extra["best_rew"], extra["best_dist"]
+        reporter.print(f"resumed from checkpoint at gen {start_gen}")
+
+    for gen in range(start_gen, cfg.general.gens):
         idx = random.choices(range(len(policies_novelties)),
                              weights=policies_novelties, k=1)[0]
         if cfg.nsr.progressive:
@@ -182,6 +208,16 @@ def main(cfg):
             np.save(path.join(archive_path, f"{gen}.np"), archive)
 
         reporter.end_gen()
+
+        if ckpt is not None:
+            ckpt.maybe_save(gen + 1, population, rs, cfg=cfg,
+                            engine=engines or None,
+                            extra={"archive": np.asarray(archive),
+                                   "policies_novelties": policies_novelties,
+                                   "policies_best_rewards": policies_best_rewards,
+                                   "time_since_best": time_since_best,
+                                   "obj_weight": obj_weight,
+                                   "best_rew": best_rew, "best_dist": best_dist})
 
 
 if __name__ == "__main__":

@@ -142,3 +142,26 @@ def test_obj_resume_from_checkpoint(tmp_path):
     cfg["general"]["name"] = "resumed"
     _run("obj.py", cfg, tmp_path)
     assert os.path.exists(tmp_path / "saved" / "CartPole-v1-resumed")
+
+
+def test_obj_ckpt_ring_resume(tmp_path):
+    """general.ckpt_every ring: a relaunch continues from the snapshot."""
+    cfg = _base(gens=2)
+    cfg["general"]["ckpt_every"] = 1
+    _run("obj.py", cfg, tmp_path)
+    ring = tmp_path / "saved" / "CartPole-v1-t" / "ckpt"
+    assert sorted(os.listdir(ring)) == ["ckpt-1.pkl", "ckpt-2.pkl"]
+    cfg["general"]["gens"] = 4
+    r = _run("obj.py", cfg, tmp_path)
+    assert "resumed from checkpoint at gen 2" in r.stdout
+    assert sorted(os.listdir(ring)) == ["ckpt-2.pkl", "ckpt-3.pkl", "ckpt-4.pkl"]
+
+
+def test_nsra_ckpt_resume(tmp_path):
+    """NSRA control state (archive, weights, selection RNG) resumes."""
+    cfg = _base(name="Hopper-v3", max_steps=30, gens=2)
+    cfg["general"]["ckpt_every"] = 2
+    _run("nsra.py", cfg, tmp_path)
+    cfg["general"]["gens"] = 3
+    r = _run("nsra.py", cfg, tmp_path)
+    assert "resumed from checkpoint at gen 2" in r.stdout
