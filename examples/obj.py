@@ -51,6 +51,15 @@ def main(cfg):
     time_since_best = 0
     noise_std_inc = 0.08
 
+    # hang watchdog: a wedged collective or kernel turns into a loud exit
+    # (code 124) restartable from the checkpoint ring, instead of silently
+    # holding the node (utils/watchdog.py)
+    wd = hb = None
+    if cfg.general.get("gen_timeout_s"):
+        from es_pytorch_amd.utils.watchdog import Heartbeat, Watchdog
+        wd = Watchdog(float(cfg.general.gen_timeout_s))
+        hb = Heartbeat(path.join("saved", full_name, "heartbeat"), comm.rank)
+
     # atomic ring checkpointing + exact resume (utils/checkpoint.py); a
     # relaunch of the same command continues bit-for-bit from the newest
     # snapshot (beyond the reference's policy-only checkpoints, obj.py:39-41)
@@ -84,8 +93,14 @@ def main(cfg):
         if cfg.policy.get("ac_std_decay", 1) != 1:
             reporter.log({"ac std": nn._action_std})
 
-        tr, gen_obstat = step_any(cfg, comm, policy, nt, env, engine, fit_fn, rs,
-                                  ranker, reporter)
+        if wd is not None:
+            with wd.guard(f"generation {gen}"):
+                tr, gen_obstat = step_any(cfg, comm, policy, nt, env, engine,
+                                          fit_fn, rs, ranker, reporter)
+            hb.beat(gen)
+        else:
+            tr, gen_obstat = step_any(cfg, comm, policy, nt, env, engine, fit_fn,
+                                      rs, ranker, reporter)
         if engine is not None:  # per-phase timers (rollout/collective/update)
             reporter.log({k: round(v, 4) for k, v in engine.timings.items()
                           if k.endswith("_s")})
