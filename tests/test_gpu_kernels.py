@@ -989,3 +989,20 @@ def test_engine_exact_checkpoint_resume(dev, tmp_path):
     np.testing.assert_array_equal(a[4].optim.m, policy_c.optim.m)
     assert a[4].optim.t == policy_c.optim.t
     assert rews_a[2:] == rews_c  # resumed generations bitwise-reproduce
+
+
+def test_twin_rank_engine_identity(dev):
+    """Two ranks sharing ONE GPU over gloo (tools/twin_rank_check.py): the
+    multi-rank engine flow — per-rank draws, fitness all-gather, redundant
+    updates — yields bitwise-identical parameters on every rank."""
+    import os
+    import subprocess
+    import sys
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    cmd = [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+           "--nproc-per-node", "2", "--standalone", "--local-addr", "127.0.0.1",
+           os.path.join(root, "tools", "twin_rank_check.py")]
+    r = subprocess.run(cmd, capture_output=True, text=True, timeout=420,
+                       env=dict(os.environ, PYTHONPATH=root))
+    assert r.returncode == 0, r.stderr[-3000:]
+    assert "TWIN-RANK OK" in r.stdout
