@@ -126,20 +126,39 @@ class RunCheckpointer:
 
     def restore(self, state: Dict[str, Any], policy: Policy,
                 rs: np.random.RandomState, cfg=None, engine=None,
-                env=None) -> Tuple[int, Dict[str, Any]]:
-        """Restore IN PLACE into the live objects; returns (next_gen, extra)."""
-        if state["world_size"] != self.comm.size:
+                env=None, allow_reshard: bool = False) -> Tuple[int, Dict[str, Any]]:
+        """Restore IN PLACE into the live objects; returns (next_gen, extra).
+
+        With ``allow_reshard=True`` a snapshot written at a different world
+        size restores too (elastic restart, e.g. an 8-GPU run resumed on 4):
+        learned state — params, moments, ObStat, schedules, archive — carries
+        over exactly, but the per-rank RNG streams must re-split, so the
+        continuation is deterministic given (snapshot, new world size) yet
+        not bitwise-identical to the uninterrupted original. Default strict
+        mode refuses the mismatch so same-size resumes stay bit-exact.
+        """
+        resharded = state["world_size"] != self.comm.size
+        if resharded and not allow_reshard:
             raise RuntimeError(
                 f"checkpoint was written by world_size={state['world_size']}, "
                 f"resuming with {self.comm.size} ranks would desync the RNG "
-                "streams — relaunch with the original rank count")
+                "streams — relaunch with the original rank count, or pass "
+                "allow_reshard=True for a deterministic (not bitwise) restart")
         saved = pickle.loads(state["policy"])
         for live, snap in zip(_as_list(policy), _as_list(saved)):
             _restore_policy(live, snap)
-        np_state, torch_state, py_state = state["rng_states"][self.comm.rank]
-        rs.set_state(np_state)
-        torch.set_rng_state(torch.from_numpy(torch_state))
-        random.setstate(py_state)
+        if resharded:
+            # fresh deterministic per-rank streams for the new rank layout
+            mix = (int(state["next_gen"]) * 1000003 + self.comm.rank * 7919
+                   + self.comm.size * 104729) & 0x7FFFFFFF
+            rs.set_state(np.random.RandomState(mix).get_state())
+            torch.manual_seed(int(state["next_gen"]))
+            random.seed(mix)
+        else:
+            np_state, torch_state, py_state = state["rng_states"][self.comm.rank]
+            rs.set_state(np_state)
+            torch.set_rng_state(torch.from_numpy(torch_state))
+            random.setstate(py_state)
         if cfg is not None and state["cfg"] is not None:
             # restore ONLY the schedule-mutated scalars: run-duration and
             # launch intent (gens, ckpt cadence, env, ...) must come from the
@@ -150,11 +169,17 @@ class RunCheckpointer:
                 sv = state["cfg"].get(sect, {})
                 if key in sv and sect in cfg and key in cfg[sect]:
                     cfg[sect][key] = sv[key]
-        ests = state.get("engine_states", [None] * self.comm.size)[self.comm.rank]
-        for e, est in zip(_as_list(engine), ests or []):
-            e.restore_from_policy(gen=state["next_gen"])
-            if est is not None:
-                e.load_checkpoint_state(est)
+        if resharded:
+            # prefetched offset draws belong to the OLD rank split; drop them
+            for e in _as_list(engine):
+                e.restore_from_policy(gen=state["next_gen"])
+        else:
+            ests = state.get("engine_states",
+                             [None] * self.comm.size)[self.comm.rank]
+            for e, est in zip(_as_list(engine), ests or []):
+                e.restore_from_policy(gen=state["next_gen"])
+                if est is not None:
+                    e.load_checkpoint_state(est)
         if env is not None and state.get("env_seed") is not None:
             env._seed = state["env_seed"]
         return state["next_gen"], dict(state["extra"])

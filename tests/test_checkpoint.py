@@ -143,3 +143,44 @@ def test_checkpoint_world2(tmp_path):
     from tests.mp_helpers import run_mp
     draws = run_mp(_ckpt_roundtrip_rank, world=2, args=(str(tmp_path / "ring"),))
     assert draws[0] != draws[1]  # streams really are per-rank
+
+
+def _ckpt_save_only_rank(rank, world, folder):
+    import torch as T
+
+    from es_pytorch_amd.nn.optimizers import Adam as A2
+    from es_pytorch_amd.parallel.comm import Comm as C2
+
+    T.manual_seed(1)
+    env = make("CartPole-v1")
+    nn = FeedForward([8], T.nn.Tanh(), env, ac_std=0.01, ob_clip=5)  # _fresh arch
+    policy = Policy(nn, 0.02, A2(len(Policy.get_flat(nn)), 0.01))
+    policy.flat_params[:] = 42.0
+    rs = np.random.RandomState(100 + rank)
+    RunCheckpointer(folder, C2(T.device("cpu")), keep=2).save(
+        9, policy, rs, extra={"tag": "reshard"})
+    return rank
+
+
+def test_elastic_reshard_restore(tmp_path):
+    """A snapshot written at world_size=2 restores into a 1-rank run with
+    allow_reshard=True: learned state carries over exactly, per-rank RNG
+    streams re-split deterministically; strict mode still refuses."""
+    from tests.mp_helpers import run_mp
+    folder = str(tmp_path / "ring")
+    run_mp(_ckpt_save_only_rank, world=2, args=(folder,))
+
+    comm, cfg, env, rs, policy, nt, _ = _fresh()  # world_size=1 here
+    ck = RunCheckpointer(folder, comm)
+    state = ck.load()
+    with pytest.raises(RuntimeError, match="allow_reshard"):
+        ck.restore(state, policy, rs)
+    next_gen, extra = ck.restore(state, policy, rs, allow_reshard=True)
+    assert next_gen == 9 and extra["tag"] == "reshard"
+    np.testing.assert_array_equal(policy.flat_params,
+                                  np.full_like(policy.flat_params, 42.0))
+    a = rs.randint(0, 10 ** 6, 4)
+    # deterministic: a second resharded restore reproduces the same stream
+    rs2 = np.random.RandomState(1)
+    ck.restore(state, policy, rs2, allow_reshard=True)
+    np.testing.assert_array_equal(rs2.randint(0, 10 ** 6, 4), a)
