@@ -72,8 +72,9 @@ def main(cfg):
                                every=ck_every)
         state = ckpt.load()
         if state is not None:
-            start_gen, extra = ckpt.restore(state, policy, rs, cfg=cfg,
-                                            engine=engine, env=env)
+            start_gen, extra = ckpt.restore(
+                state, policy, rs, cfg=cfg, engine=engine, env=env,
+                allow_reshard=bool(cfg.general.get("ckpt_allow_reshard", False)))
             best_max_rew = extra.get("best_max_rew", best_max_rew)
             time_since_best = extra.get("time_since_best", time_since_best)
             if 0 < elite < 1:
