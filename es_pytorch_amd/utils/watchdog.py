@@ -125,3 +125,25 @@ class Heartbeat:
         """Beats older than ``timeout_s`` — ranks that stopped progressing."""
         now = time.time()
         return [b for b in Heartbeat.read(folder) if now - b["ts"] > timeout_s]
+
+
+def _main(argv=None):
+    import argparse
+
+    p = argparse.ArgumentParser(description="heartbeat liveness check")
+    p.add_argument("folder", help="heartbeat dir, e.g. saved/<run>/heartbeat")
+    p.add_argument("--timeout", type=float, default=300.0,
+                   help="seconds without a beat before a rank counts as stalled")
+    args = p.parse_args(argv)
+    beats = Heartbeat.read(args.folder)
+    stalled = Heartbeat.stalled_ranks(args.folder, args.timeout)
+    now = time.time()
+    for b in beats:
+        mark = "STALLED" if b in stalled else "ok"
+        print(f"rank {b['rank']}: gen {b['gen']}, last beat {now - b['ts']:.0f}s "
+              f"ago [{mark}]")
+    return 1 if stalled else 0
+
+
+if __name__ == "__main__":
+    sys.exit(_main())

@@ -55,3 +55,11 @@ def test_watchdog_default_action_exits_124():
     assert r.returncode == 124
     assert "watchdog" in r.stderr and "hung gen" in r.stderr
     assert "Current thread" in r.stderr or "Thread" in r.stderr  # stack dump
+
+
+def test_heartbeat_cli(tmp_path, capsys):
+    from es_pytorch_amd.utils.watchdog import _main
+    Heartbeat(str(tmp_path), 0).beat(gen=12)
+    assert _main([str(tmp_path), "--timeout", "300"]) == 0
+    assert "rank 0: gen 12" in capsys.readouterr().out
+    assert _main([str(tmp_path), "--timeout", "0"]) == 1  # everything stalled
