@@ -78,7 +78,8 @@ class GpuEngine:
                  rs: np.random.RandomState, objective: str = "reward",
                  use_graph: bool = True, novelty_k: int = 10,
                  fused: Optional[bool] = None, rollout_mode: str = "step",
-                 split_dyn: Optional[bool] = None):
+                 split_dyn: Optional[bool] = None,
+                 pair_rollout: Optional[bool] = None):
         # rollout_mode: "step" = one kernel per env step for the population
         # (graph-replayed); "episode" = ONE kernel per generation, each block
         # runs its member's whole episode (members are mutually independent,
@@ -200,6 +201,26 @@ class GpuEngine:
         if self.split_dyn:
             self.act_scratch = torch.empty(self.B * 64, dtype=torch.float32, device=d)
 
+        # antithetic-pair rollout: the +/- members of a pair share one HBM
+        # sigma*eps stream plus the L2-resident shared theta stream
+        # (mlp_layers_pair) — per-step HBM weight traffic halves vs the
+        # materialized per-member blobs. Effective weights become
+        # bf16(theta) +- bf16(sigma*eps) (two bf16 roundings instead of one).
+        want_pair = bool(cfg.general.get("pair_rollout", False)) if pair_rollout \
+            is None else bool(pair_rollout)
+        self.pair_rollout = (want_pair and self.fused and not self.split_dyn
+                             and self.rollout_mode == "step"
+                             and self.steps_per_launch == 1 and self.pairs >= 1)
+        if self.pair_rollout:
+            self.theta_row = torch.empty((1, self.row_stride), dtype=torch.bfloat16,
+                                         device=d)
+            self.eps_rows = torch.empty((self.pairs, self.row_stride),
+                                        dtype=torch.bfloat16, device=d)
+            self._zeros_n = torch.zeros(self.n, dtype=torch.float32, device=d)
+            self._zero_off = torch.zeros(1, dtype=torch.int64, device=d)
+            self._zero_sign = torch.zeros(1, dtype=torch.float32, device=d)
+            self._one_signs = torch.ones(self.pairs, dtype=torch.float32, device=d)
+
     # ------------------------------------------------------------------ ops
     def _stream(self):
         return torch.cuda.current_stream(self.device).cuda_stream if \
@@ -207,6 +228,20 @@ class GpuEngine:
 
     def _pheno(self):
         std = float(self.policy.std)
+        if self.pair_rollout:
+            # theta row: sign 0 -> bf16(theta + 0*noise) via the same kernel
+            ops.check(ops.hip().es_pheno_bf16(
+                self.theta_row.data_ptr(), self.theta.data_ptr(),
+                self.nt.noise.data_ptr(), self._zero_off.data_ptr(),
+                self._zero_sign.data_ptr(), 1, self.n, self.row_stride, 0.0,
+                self._stream()), "es_pheno_bf16")
+            # sigma*eps rows: zero theta + sign +1 -> bf16(sigma*noise[off_p:])
+            ops.check(ops.hip().es_pheno_bf16(
+                self.eps_rows.data_ptr(), self._zeros_n.data_ptr(),
+                self.nt.noise.data_ptr(), self.offsets.data_ptr(),
+                self._one_signs.data_ptr(), self.pairs, self.n, self.row_stride,
+                std, self._stream()), "es_pheno_bf16")
+            return
         ops.check(ops.hip().es_pheno_bf16(
             self.weights.data_ptr(), self.theta.data_ptr(), self.nt.noise.data_ptr(),
             self.offsets.data_ptr(), self.signs.data_ptr(), self.M, self.n,
@@ -231,6 +266,28 @@ class GpuEngine:
         main grid stays an exact multiple of the CU slot count."""
         env = self.env
         goal_ptr = env.goal.data_ptr() if env.goal_conditioned else None
+        if self.pair_rollout:
+            ops.check(ops.hip().es_loco_pair_step(
+                self.theta_row.data_ptr(), self.eps_rows.data_ptr(),
+                self.obmean.data_ptr(), self.obstd.data_ptr(),
+                self.dims_arr.ctypes.data, len(self.dims_arr),
+                self.seed_dev.data_ptr(), t + 1,
+                float(self.policy._module.ob_clip), self.acstd_dev.data_ptr(),
+                self.row_stride,
+                env.s.data_ptr(), env.pos.data_ptr(), goal_ptr,
+                env.A_bf16.data_ptr(), env.B.data_ptr(), env.b0.data_ptr(),
+                env.wv.data_ptr(), env.wa.data_ptr(), env.wy.data_ptr(),
+                env.wh.data_ptr(),
+                self.alive.data_ptr(), self.rew_total.data_ptr(),
+                self.member_steps.data_ptr(), self.behv.data_ptr(),
+                self.mo_sum.data_ptr(), self.mo_sumsq.data_ptr(),
+                self.pairs, env.sdim, env.ac_dim, int(env.goal_conditioned),
+                int(env.terminate_on_fall), (self.M - 1) * self.eps, self.bins,
+                self.eps, self.act_mode,
+                float(env.leak), float(env.ctrl_cost), float(env.alive_bonus),
+                float(env.fall_threshold), float(env.dt), self._stream()),
+                "es_loco_pair_step")
+            return
         common = (
             self.weights.data_ptr(), self.obmean.data_ptr(), self.obstd.data_ptr(),
             self.dims_arr.ctypes.data, len(self.dims_arr), self.seed_dev.data_ptr(),
@@ -255,13 +312,17 @@ class GpuEngine:
             ops.check(ops.hip().es_loco_step(*common, self._stream()), "es_loco_step")
 
     def _loco_episode(self, member_base: int, n_members: int, noiseless_from: int,
-                      n_steps: Optional[int] = None, salt_base: int = 0):
+                      n_steps: Optional[int] = None, salt_base: int = 0,
+                      weights_ptr: Optional[int] = None, wrow0: int = 0):
         """n_steps consecutive env steps (default: the whole episode) for
-        [member_base, member_base+n_members) slots in one launch."""
+        [member_base, member_base+n_members) slots in one launch. weights_ptr
+        + wrow0 let the call run on a sub-blob (pair mode: the 1-row theta
+        blob serves the noiseless slots, whose weights row is wrow0)."""
         env = self.env
         goal_ptr = env.goal.data_ptr() if env.goal_conditioned else None
         ops.check(ops.hip().es_loco_episode(
-            self.weights.data_ptr(), self.obmean.data_ptr(), self.obstd.data_ptr(),
+            self.weights.data_ptr() if weights_ptr is None else weights_ptr,
+            self.obmean.data_ptr(), self.obstd.data_ptr(),
             self.dims_arr.ctypes.data, len(self.dims_arr), self.seed_dev.data_ptr(),
             self.max_steps if n_steps is None else n_steps,
             float(self.policy._module.ob_clip),
@@ -272,7 +333,7 @@ class GpuEngine:
             self.alive.data_ptr(), self.rew_total.data_ptr(),
             self.member_steps.data_ptr(), self.behv.data_ptr(),
             self.mo_sum.data_ptr(), self.mo_sumsq.data_ptr(),
-            member_base, n_members, salt_base, env.sdim, env.ac_dim,
+            member_base, n_members, salt_base, wrow0, env.sdim, env.ac_dim,
             int(env.goal_conditioned), int(env.terminate_on_fall), noiseless_from,
             self.bins, self.eps, self.act_mode,
             float(env.leak), float(env.ctrl_cost), float(env.alive_bonus),
@@ -280,7 +341,12 @@ class GpuEngine:
             "es_loco_episode")
 
     def _loco_noiseless_episode(self):
-        self._loco_episode((self.M - 1) * self.eps, self.eps, 0)
+        if self.pair_rollout:
+            self._loco_episode((self.M - 1) * self.eps, self.eps, 0,
+                               weights_ptr=self.theta_row.data_ptr(),
+                               wrow0=self.M - 1)
+        else:
+            self._loco_episode((self.M - 1) * self.eps, self.eps, 0)
 
     # ------------------------------------------------------------- rollout
     def _step_body(self, t: int):

@@ -195,3 +195,172 @@ __device__ __forceinline__ float es_actnoise(uint64_t seed, uint64_t ctr) {
   esrng::f32x4 v = esrng::normal4(ctr, seed, 0xACu);
   return v.x;
 }
+
+// ---- antithetic-pair forward -----------------------------------------------
+// ES structure exploit: the +noise and -noise members of a pair share the
+// SAME perturbation row. Instead of streaming two materialized bf16 blobs
+// bf16(theta+sigma*eps) and bf16(theta-sigma*eps) from HBM (2 x n bytes per
+// pair per step — the measured dominant cost of the whole framework), the
+// pair forward streams ONE bf16(sigma*eps) row (HBM, nt loads) plus the
+// bf16(theta) row that every pair shares (regular loads -> L2-resident
+// across the grid), and forms W+/W- in registers. HBM weight traffic per
+// step halves; effective weights are bf16(theta) +- bf16(sigma*eps), i.e.
+// two bf16 roundings instead of one (parity tests use a matching torch
+// reference).
+__device__ __forceinline__ void bf8_fma_pair(uint4 t, uint4 e, float xp, float xm,
+                                             float* accp, float* accm) {
+  const uint32_t ts[4] = {t.x, t.y, t.z, t.w};
+  const uint32_t es_[4] = {e.x, e.y, e.z, e.w};
+#pragma unroll
+  for (int q = 0; q < 4; ++q) {
+    const float t0 = bf2f((uint16_t)(ts[q] & 0xFFFFu));
+    const float e0 = bf2f((uint16_t)(es_[q] & 0xFFFFu));
+    const float t1 = bf2f((uint16_t)(ts[q] >> 16));
+    const float e1 = bf2f((uint16_t)(es_[q] >> 16));
+    accp[2 * q] = fmaf(t0 + e0, xp, accp[2 * q]);
+    accm[2 * q] = fmaf(t0 - e0, xm, accm[2 * q]);
+    accp[2 * q + 1] = fmaf(t1 + e1, xp, accp[2 * q + 1]);
+    accm[2 * q + 1] = fmaf(t1 - e1, xm, accm[2 * q + 1]);
+  }
+}
+
+// Dual-member mlp_layers: one theta stream + one sigma*eps stream drive BOTH
+// members' forwards. Same (oi, ip) tiling, depth-4 pipelining and partial
+// layout as mlp_layers, with per-member accumulator/partial sets. `partial`
+// needs 2 x 256*8 floats; bufAp/bufAm/bufBp/bufBm are maxdim floats each.
+// Returns the +/- output pointers via xp_out/xm_out.
+__device__ __forceinline__ void mlp_layers_pair(
+    const uint16_t* __restrict__ tb, const uint16_t* __restrict__ eb,
+    const MlpShape& sh, float* bufAp, float* bufAm, float* bufBp, float* bufBm,
+    float* partial, int tid, int nthreads, int act_final,
+    float** xp_out, float** xm_out) {
+  float* xp = bufAp;
+  float* xm = bufAm;
+  float* yp = bufBp;
+  float* ym = bufBm;
+  float* partm = partial + 256 * 8;
+  for (int l = 0; l < sh.n_layers; ++l) {
+    const int I = sh.dims[l], O = sh.dims[l + 1];
+    const uint16_t* Tt = tb + sh.woff[l];
+    const uint16_t* Et = eb + sh.woff[l];
+    const uint16_t* TBs = tb + sh.boff[l];
+    const uint16_t* EBs = eb + sh.boff[l];
+    const bool do_act = (l < sh.n_layers - 1) || act_final;
+
+    if (sh.vec_ok[l]) {
+      const int OCT = O >> 3;
+      const int PART = nthreads / OCT;
+      const int oi = tid % OCT, ip = tid / OCT;
+      float accp[8], accm[8];
+#pragma unroll
+      for (int q = 0; q < 8; ++q) accp[q] = accm[q] = 0.0f;
+      if (ip < PART) {
+        const uint16_t* tcol = Tt + (oi << 3);
+        const uint16_t* ecol = Et + (oi << 3);
+        typedef uint32_t u32x4v __attribute__((ext_vector_type(4)));
+        // eps rows are touched by exactly one block -> nt; theta is shared
+        // by the whole grid -> regular load, stays L2/MALL-resident
+        auto lde = [&](int i) {
+          u32x4v v = __builtin_nontemporal_load(
+              reinterpret_cast<const u32x4v*>(ecol + (int64_t)i * O));
+          uint4 w;
+          w.x = v.x; w.y = v.y; w.z = v.z; w.w = v.w;
+          return w;
+        };
+        auto ldt = [&](int i) {
+          return *reinterpret_cast<const uint4*>(tcol + (int64_t)i * O);
+        };
+        int i = ip;
+        const int step4 = PART * 4;
+        if (i + 3 * PART < I) {
+          uint4 t0 = ldt(i), t1 = ldt(i + PART), t2 = ldt(i + 2 * PART),
+                t3 = ldt(i + 3 * PART);
+          uint4 e0 = lde(i), e1 = lde(i + PART), e2 = lde(i + 2 * PART),
+                e3 = lde(i + 3 * PART);
+          for (; i + 7 * PART < I; i += step4) {
+            const uint4 nt0 = ldt(i + 4 * PART), nt1 = ldt(i + 5 * PART),
+                        nt2 = ldt(i + 6 * PART), nt3 = ldt(i + 7 * PART);
+            const uint4 ne0 = lde(i + 4 * PART), ne1 = lde(i + 5 * PART),
+                        ne2 = lde(i + 6 * PART), ne3 = lde(i + 7 * PART);
+            bf8_fma_pair(t0, e0, xp[i], xm[i], accp, accm);
+            bf8_fma_pair(t1, e1, xp[i + PART], xm[i + PART], accp, accm);
+            bf8_fma_pair(t2, e2, xp[i + 2 * PART], xm[i + 2 * PART], accp, accm);
+            bf8_fma_pair(t3, e3, xp[i + 3 * PART], xm[i + 3 * PART], accp, accm);
+            t0 = nt0; t1 = nt1; t2 = nt2; t3 = nt3;
+            e0 = ne0; e1 = ne1; e2 = ne2; e3 = ne3;
+          }
+          bf8_fma_pair(t0, e0, xp[i], xm[i], accp, accm);
+          bf8_fma_pair(t1, e1, xp[i + PART], xm[i + PART], accp, accm);
+          bf8_fma_pair(t2, e2, xp[i + 2 * PART], xm[i + 2 * PART], accp, accm);
+          bf8_fma_pair(t3, e3, xp[i + 3 * PART], xm[i + 3 * PART], accp, accm);
+          i += step4;
+        }
+        for (; i < I; i += PART)
+          bf8_fma_pair(ldt(i), lde(i), xp[i], xm[i], accp, accm);
+#pragma unroll
+        for (int q = 0; q < 8; ++q) {
+          partial[(ip * OCT + oi) * 8 + q] = accp[q];
+          partm[(ip * OCT + oi) * 8 + q] = accm[q];
+        }
+      }
+      __syncthreads();
+      for (int o = tid; o < O; o += nthreads) {
+        const float tB = bf2f(TBs[o]), eB = bf2f(EBs[o]);
+        float sp = tB + eB, sm = tB - eB;
+        const int oo = o >> 3, j = o & 7;
+        for (int p = 0; p < PART; ++p) {
+          sp += partial[(p * OCT + oo) * 8 + j];
+          sm += partm[(p * OCT + oo) * 8 + j];
+        }
+        yp[o] = do_act ? tanhf(sp) : sp;
+        ym[o] = do_act ? tanhf(sm) : sm;
+      }
+    } else {
+      // dual tiled scalar path (small / odd layers, e.g. the action head)
+      const int PART = nthreads / O;
+      if (PART > 1) {
+        const int oi = tid % O, ip = tid / O;
+        float accp = 0.0f, accm = 0.0f;
+        if (ip < PART) {
+          for (int i = ip; i < I; i += PART) {
+            const float tw = bf2f(Tt[(int64_t)i * O + oi]);
+            const float ew = bf2f(Et[(int64_t)i * O + oi]);
+            accp = fmaf(tw + ew, xp[i], accp);
+            accm = fmaf(tw - ew, xm[i], accm);
+          }
+          partial[ip * O + oi] = accp;
+          partm[ip * O + oi] = accm;
+        }
+        __syncthreads();
+        for (int o = tid; o < O; o += nthreads) {
+          const float tB = bf2f(TBs[o]), eB = bf2f(EBs[o]);
+          float sp = tB + eB, sm = tB - eB;
+          for (int p = 0; p < PART; ++p) {
+            sp += partial[p * O + o];
+            sm += partm[p * O + o];
+          }
+          yp[o] = do_act ? tanhf(sp) : sp;
+          ym[o] = do_act ? tanhf(sm) : sm;
+        }
+      } else {
+        for (int o = tid; o < O; o += nthreads) {
+          const float tB = bf2f(TBs[o]), eB = bf2f(EBs[o]);
+          float accp = tB + eB, accm = tB - eB;
+          for (int i = 0; i < I; ++i) {
+            const float tw = bf2f(Tt[(int64_t)i * O + o]);
+            const float ew = bf2f(Et[(int64_t)i * O + o]);
+            accp = fmaf(tw + ew, xp[i], accp);
+            accm = fmaf(tw - ew, xm[i], accm);
+          }
+          yp[o] = do_act ? tanhf(accp) : accp;
+          ym[o] = do_act ? tanhf(accm) : accm;
+        }
+      }
+    }
+    __syncthreads();
+    float* t = xp; xp = yp; yp = t;
+    t = xm; xm = ym; ym = t;
+  }
+  *xp_out = xp;
+  *xm_out = xm;
+}

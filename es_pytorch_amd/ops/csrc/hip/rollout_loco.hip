@@ -29,6 +29,8 @@ struct LocoArgs {
   int bins;            // >1: K9 binned-action decode (FFBinned)
   int act_mode;        // 0 plain, 1 binned, 2 integ-gauss, 3 integ-gauss-multi
   int eps;             // episodes per perturbation: slot b uses weights row b/eps
+  int wrow0;           // weights-row index of slot 0 (episode kernel on a
+                       // sub-blob, e.g. the pair path's 1-row theta blob)
   float leak, ctrl, alive_bonus, fall_thr, dt, ob_clip;
   int64_t row_stride;
 };
@@ -43,15 +45,12 @@ struct LocoPtrs {
   float *alive, *rew_total, *member_steps, *behv, *mo_sum, *mo_sumsq;
 };
 
-__device__ __forceinline__ void loco_fwd_body(
-    const MlpShape& sh, const LocoArgs& la, const LocoPtrs& P, int b, uint64_t salt,
-    float* bufA, float* bufB, float* partial, float* raws, float* abuf) {
-  const int tid = threadIdx.x;
-  const int nth = blockDim.x;
-  const int S = la.S, A = la.A;
+// ---- build normalized obs for slot b; keep raw state in LDS --------------
+__device__ __forceinline__ void loco_build_obs(
+    const LocoArgs& la, const LocoPtrs& P, int b, float* bufA, float* raws,
+    int tid, int nth) {
+  const int S = la.S;
   const float* sb = P.s_glob + (int64_t)b * S;
-
-  // ---- build normalized obs; keep raw state in LDS -----------------------
   for (int i = tid; i < S; i += nth) {
     const float v = sb[i];
     raws[i] = v;
@@ -62,11 +61,13 @@ __device__ __forceinline__ void loco_fwd_body(
     bufA[S + tid] = fclampf((rel - P.obmean[S + tid]) / P.obstd[S + tid], -la.ob_clip,
                             la.ob_clip);
   }
-  __syncthreads();
+}
 
-  // ---- policy forward ----------------------------------------------------
-  const uint16_t* wb = P.weights + (int64_t)(b / la.eps) * la.row_stride;
-  const float* aout = mlp_layers(wb, sh, bufA, bufB, partial, tid, nth, 1);
+// ---- net output -> clamped env action for slot b (modes 0/1/2/3) ---------
+__device__ __forceinline__ void loco_decode_action(
+    const MlpShape& sh, const LocoArgs& la, const LocoPtrs& P, int b, uint64_t salt,
+    const float* aout, float* abuf, int tid) {
+  const int A = la.A;
   const uint64_t seed = P.seed_dev ? (*P.seed_dev + salt) : salt;
   const float ac_std = P.ac_std_dev ? *P.ac_std_dev : 0.0f;
   if (la.act_mode == 2 || la.act_mode == 3) {
@@ -95,6 +96,18 @@ __device__ __forceinline__ void loco_fwd_body(
       a += ac_std * es_actnoise(seed, (uint64_t)b * A + tid);
     abuf[tid] = fclampf(a, -1.0f, 1.0f);  // env action clamp (locomotion.py)
   }
+}
+
+__device__ __forceinline__ void loco_fwd_body(
+    const MlpShape& sh, const LocoArgs& la, const LocoPtrs& P, int b, uint64_t salt,
+    float* bufA, float* bufB, float* partial, float* raws, float* abuf) {
+  const int tid = threadIdx.x;
+  const int nth = blockDim.x;
+  loco_build_obs(la, P, b, bufA, raws, tid, nth);
+  __syncthreads();
+  const uint16_t* wb = P.weights + (int64_t)(b / la.eps - la.wrow0) * la.row_stride;
+  const float* aout = mlp_layers(wb, sh, bufA, bufB, partial, tid, nth, 1);
+  loco_decode_action(sh, la, P, b, salt, aout, abuf, tid);
   __syncthreads();
 }
 
@@ -391,6 +404,106 @@ static int loco_launch_dyn(const LocoArgs& la, const LocoPtrs& P, const float* a
   return 0;
 }
 
+// ---- antithetic-pair rollout step ------------------------------------------
+// One block per (pair, episode): the +noise and -noise members' forwards run
+// together off ONE HBM sigma*eps stream plus the L2-resident shared theta
+// stream (mlp_layers_pair, mlp_core.h) — the per-step HBM weight traffic
+// halves vs materialized per-member blobs. Dynamics shares each A octet
+// between the two members; the per-member epilogue is loco_dyn_finish
+// verbatim, so per-slot bookkeeping is identical to the fused step.
+// Effective weights are bf16(theta) +- bf16(sigma*eps) (two roundings); with
+// sigma = 0 the trajectories are BITWISE-identical to es_loco_step.
+__device__ __forceinline__ void loco_dyn_partials_pair(
+    const uint16_t* Am, int S, const float* rawsP, const float* rawsM,
+    float* partP, float* partM, int tid, int nth) {
+  const int OCT = S >> 3;
+  const int PART = nth / OCT;
+  const int oi = tid % OCT, ip = tid / OCT;
+  float accp[8], accm[8];
+#pragma unroll
+  for (int q = 0; q < 8; ++q) accp[q] = accm[q] = 0.0f;
+  if (ip < PART) {
+    const uint16_t* acol = Am + (oi << 3);
+    auto ld = [&](int i) {
+      return *reinterpret_cast<const uint4*>(acol + (int64_t)i * S);
+    };
+    auto fma2 = [&](const uint4& c, int i) {
+      bf8_fma(c, rawsP[i], accp);
+      bf8_fma(c, rawsM[i], accm);
+    };
+    int i = ip;
+    const int step4 = PART * 4;
+    if (i + 3 * PART < S) {
+      uint4 c0 = ld(i), c1 = ld(i + PART), c2 = ld(i + 2 * PART), c3 = ld(i + 3 * PART);
+      for (; i + 7 * PART < S; i += step4) {
+        const uint4 n0 = ld(i + 4 * PART), n1 = ld(i + 5 * PART),
+                    n2 = ld(i + 6 * PART), n3 = ld(i + 7 * PART);
+        fma2(c0, i);
+        fma2(c1, i + PART);
+        fma2(c2, i + 2 * PART);
+        fma2(c3, i + 3 * PART);
+        c0 = n0; c1 = n1; c2 = n2; c3 = n3;
+      }
+      fma2(c0, i);
+      fma2(c1, i + PART);
+      fma2(c2, i + 2 * PART);
+      fma2(c3, i + 3 * PART);
+      i += step4;
+    }
+    for (; i < S; i += PART) fma2(ld(i), i);
+#pragma unroll
+    for (int q = 0; q < 8; ++q) {
+      partP[(ip * OCT + oi) * 8 + q] = accp[q];
+      partM[(ip * OCT + oi) * 8 + q] = accm[q];
+    }
+  }
+  __syncthreads();
+}
+
+__device__ __forceinline__ void loco_pair_step_body(
+    const MlpShape& sh, const LocoArgs& la, const LocoPtrs& P,
+    const uint16_t* tb, const uint16_t* eb, int bp, int bm, uint64_t salt,
+    float* bufAp, float* bufAm, float* bufBp, float* bufBm, float* partial,
+    float* rawsP, float* rawsM, float* abufP, float* abufM) {
+  const int tid = threadIdx.x, nth = blockDim.x;
+  loco_build_obs(la, P, bp, bufAp, rawsP, tid, nth);
+  loco_build_obs(la, P, bm, bufAm, rawsM, tid, nth);
+  __syncthreads();
+  float *ap, *am;
+  mlp_layers_pair(tb, eb, sh, bufAp, bufAm, bufBp, bufBm, partial, tid, nth, 1,
+                  &ap, &am);
+  loco_decode_action(sh, la, P, bp, salt, ap, abufP, tid);
+  loco_decode_action(sh, la, P, bm, salt, am, abufM, tid);
+  __syncthreads();
+  if (la.S % 8 == 0)
+    loco_dyn_partials_pair(P.Am, la.S, rawsP, rawsM, partial, partial + 2048,
+                           tid, nth);
+  loco_dyn_finish(la, P, bp, rawsP, abufP, partial, tid, nth);
+  loco_dyn_finish(la, P, bm, rawsM, abufM, partial + 2048, tid, nth);
+}
+
+__global__ void __launch_bounds__(256)
+loco_pair_step_kernel(MlpShape sh, LocoArgs la, LocoPtrs P, const uint16_t* tb,
+                      const uint16_t* eb, int n_pairs, uint64_t salt) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* bufAp = reinterpret_cast<float*>(smem);
+  float* bufAm = bufAp + sh.maxdim;
+  float* bufBp = bufAm + sh.maxdim;
+  float* bufBm = bufBp + sh.maxdim;
+  float* partial = bufBm + sh.maxdim;  // 2 x 256*8 (forward, then dynamics)
+  float* rawsP = partial + 2 * 256 * 8;
+  float* rawsM = rawsP + ((la.S + 3) & ~3);
+  float* abufP = rawsM + ((la.S + 3) & ~3);
+  float* abufM = abufP + 64;
+  const int q = blockIdx.x;
+  const int p = q / la.eps, e = q % la.eps;
+  const int bp = p * la.eps + e;
+  const int bm = (n_pairs + p) * la.eps + e;
+  loco_pair_step_body(sh, la, P, tb, eb + (int64_t)p * la.row_stride, bp, bm, salt,
+                      bufAp, bufAm, bufBp, bufBm, partial, rawsP, rawsM, abufP,
+                      abufM);
+}
+
 static int loco_prepare(MlpShape* sh, LocoArgs* la, const int32_t* dims_host, int32_t ndims,
                         int64_t row_stride, float ob_clip, int32_t sdim, int32_t adim,
                         int32_t goal_flag, int32_t terminate, int32_t noiseless_from,
@@ -410,6 +523,7 @@ static int loco_prepare(MlpShape* sh, LocoArgs* la, const int32_t* dims_host, in
   la->terminate = terminate; la->noiseless_from = noiseless_from; la->bins = bins;
   la->act_mode = act_mode;
   la->eps = eps > 0 ? eps : 1;
+  la->wrow0 = 0;
   la->leak = leak; la->ctrl = ctrl; la->alive_bonus = alive_bonus; la->fall_thr = fall_thr;
   la->dt = dt; la->ob_clip = ob_clip; la->row_stride = row_stride;
   *lds = (unsigned)(mlp_lds_bytes(sh->maxdim) + (((sdim + 3) & ~3) + 64 + 8) * 4);
@@ -502,6 +616,40 @@ extern "C" int es_loco_step_split(
   }
 }
 
+// Antithetic-pair step: theta_row = (1, row_stride) bf16(theta);
+// eps_rows = (n_pairs, row_stride) bf16(sigma*eps). Grid = n_pairs * eps
+// blocks, each evaluating the +/- slots of one (pair, episode).
+extern "C" int es_loco_pair_step(
+    const void* theta_row, const void* eps_rows, const void* obmean,
+    const void* obstd, const int32_t* dims_host, int32_t ndims, const void* seed_dev,
+    uint64_t salt, float ob_clip, const void* ac_std_dev, int64_t row_stride,
+    void* s_glob, void* pos, const void* goal, const void* Am, const void* Bm,
+    const void* b0, const void* wv, const void* wa, const void* wy, const void* wh,
+    void* alive, void* rew_total, void* member_steps, void* behv, void* mo_sum,
+    void* mo_sumsq, int32_t n_pairs, int32_t sdim, int32_t adim, int32_t goal_flag,
+    int32_t terminate, int32_t noiseless_from, int32_t bins, int32_t eps,
+    int32_t act_mode, float leak, float ctrl, float alive_bonus, float fall_thr,
+    float dt, void* stream) {
+  MlpShape sh;
+  LocoArgs la;
+  unsigned lds_unused;
+  int rc = loco_prepare(&sh, &la, dims_host, ndims, row_stride, ob_clip, sdim, adim,
+                        goal_flag, terminate, noiseless_from, bins, eps, act_mode, leak,
+                        ctrl, alive_bonus, fall_thr, dt, &lds_unused);
+  if (rc) return rc;
+  LocoPtrs P = loco_ptrs(nullptr, obmean, obstd, ac_std_dev, seed_dev, s_glob, pos,
+                         goal, Am, Bm, b0, wv, wa, wy, wh, alive, rew_total,
+                         member_steps, behv, mo_sum, mo_sumsq);
+  const int Spad = (sdim + 3) & ~3;
+  const unsigned lds =
+      (unsigned)((4 * sh.maxdim + 2 * 256 * 8 + 2 * Spad + 2 * 64) * sizeof(float));
+  const unsigned grid = (unsigned)(n_pairs * la.eps);
+  loco_pair_step_kernel<<<dim3(grid), dim3(256), lds, (hipStream_t)stream>>>(
+      sh, la, P, (const uint16_t*)theta_row, (const uint16_t*)eps_rows, n_pairs, salt);
+  ES_CHECK_LAUNCH();
+  return 0;
+}
+
 extern "C" int es_loco_episode(const void* weights, const void* obmean, const void* obstd,
                                const int32_t* dims_host, int32_t ndims, const void* seed_dev,
                                int32_t n_steps, float ob_clip, const void* ac_std_dev,
@@ -511,8 +659,8 @@ extern "C" int es_loco_episode(const void* weights, const void* obmean, const vo
                                const void* wa, const void* wy, const void* wh, void* alive,
                                void* rew_total, void* member_steps, void* behv,
                                void* mo_sum, void* mo_sumsq, int32_t member_base,
-                               int32_t n_members, int32_t salt_base, int32_t sdim,
-                               int32_t adim,
+                               int32_t n_members, int32_t salt_base, int32_t wrow0,
+                               int32_t sdim, int32_t adim,
                                int32_t goal_flag, int32_t terminate, int32_t noiseless_from,
                                int32_t bins, int32_t eps, int32_t act_mode, float leak,
                                float ctrl, float alive_bonus, float fall_thr, float dt,
@@ -524,6 +672,7 @@ extern "C" int es_loco_episode(const void* weights, const void* obmean, const vo
                         goal_flag, terminate, noiseless_from, bins, eps, act_mode, leak,
                         ctrl, alive_bonus, fall_thr, dt, &lds);
   if (rc) return rc;
+  la.wrow0 = wrow0;
   LocoPtrs P = loco_ptrs(weights, obmean, obstd, ac_std_dev, seed_dev, s_glob, pos, goal,
                          Am, Bm, b0, wv, wa, wy, wh, alive, rew_total, member_steps, behv,
                          mo_sum, mo_sumsq);

@@ -1006,3 +1006,64 @@ def test_twin_rank_engine_identity(dev):
                        env=dict(os.environ, PYTHONPATH=root))
     assert r.returncode == 0, r.stderr[-3000:]
     assert "TWIN-RANK OK" in r.stdout
+
+
+def _pair_engine_fits(dev, std, pair, seed=92, gens=2, max_steps=20, ppg=8):
+    import numpy as np
+    from es_pytorch_amd.config import AttrDict
+    from es_pytorch_amd.core.engine import GpuEngine
+    from es_pytorch_amd.core.noisetable import NoiseTable
+    from es_pytorch_amd.core.policy import Policy
+    from es_pytorch_amd.envs import make_batched
+    from es_pytorch_amd.nn.nn import FeedForward
+    from es_pytorch_amd.nn.optimizers import Adam
+    from es_pytorch_amd.parallel.comm import Comm
+    from es_pytorch_amd.utils.rankers import CenteredRanker
+
+    torch.manual_seed(seed)
+    comm = Comm(dev)
+    cfg = AttrDict({"env": {"name": "Humanoid-v2", "max_steps": max_steps},
+                    "noise": {"tbl_size": 1_000_000, "std": std},
+                    "policy": {"layer_sizes": [64], "ac_std": 0.01, "l2coeff": 0.005,
+                               "lr": 0.01, "ob_clip": 5, "save_obs_chance": 1.0},
+                    "general": {"policies_per_gen": ppg, "batch_size": 500,
+                                "seed": 2}})
+    env = make_batched("Humanoid-v2", ppg + 1, dev, max_steps=max_steps,
+                       terminate_on_fall=True)
+    nn = FeedForward([64], torch.nn.Tanh(), env, 0.01, 5)
+    policy = Policy(nn, std, Adam(len(Policy.get_flat(nn)), 0.01))
+    nt = NoiseTable.create_shared(comm, 1_000_000, len(policy), seed=3, device=dev)
+    rs = np.random.RandomState(seed + 1)
+    eng = GpuEngine(cfg, comm, policy, nt, env, rs, use_graph=True,
+                    pair_rollout=pair)
+    assert eng.pair_rollout == pair
+    ranker = CenteredRanker()
+    for _ in range(gens):
+        eng.step(ranker)
+    torch.cuda.synchronize(dev)
+    fits = np.concatenate([ranker.fits_pos, ranker.fits_neg]).ravel()
+    return fits, eng.theta.cpu().clone(), eng.rew_total.cpu().clone()
+
+
+def test_pair_rollout_sigma0_bitwise(dev):
+    """sigma=0: pair effective weights bf16(theta)+-bf16(0) == the fused
+    path's bf16(theta+0), so the whole pipeline (slot mapping, obs, decode,
+    shared-A dynamics, epilogue) must match BITWISE."""
+    import numpy as np
+    a = _pair_engine_fits(dev, std=0.0, pair=False)
+    b = _pair_engine_fits(dev, std=0.0, pair=True)
+    np.testing.assert_array_equal(a[0], b[0])
+    assert torch.equal(a[1], b[1]) and torch.equal(a[2], b[2])
+
+
+def test_pair_rollout_sigma_close_and_deterministic(dev):
+    """sigma>0: pair weights round twice (bf16(theta)+-bf16(sigma*eps)) vs
+    once — short-horizon fitnesses stay close; the pair path is bitwise
+    repeat-deterministic."""
+    import numpy as np
+    a = _pair_engine_fits(dev, std=0.02, pair=False, gens=1, max_steps=10)
+    b = _pair_engine_fits(dev, std=0.02, pair=True, gens=1, max_steps=10)
+    np.testing.assert_allclose(a[0], b[0], rtol=0.05, atol=0.5)
+    c = _pair_engine_fits(dev, std=0.02, pair=True, gens=1, max_steps=10)
+    np.testing.assert_array_equal(b[0], c[0])
+    assert torch.equal(b[1], c[1])

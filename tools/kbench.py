@@ -28,6 +28,7 @@ def main():
     p.add_argument("--episode", action="store_true", help="whole-episode rollout mode")
     p.add_argument("--chunk", type=int, default=1, help="env steps per launch")
     p.add_argument("--split", action="store_true", help="split-dynamics rollout")
+    p.add_argument("--pair", action="store_true", help="antithetic-pair rollout")
     p.add_argument("--dyn-group", type=int, default=5,
                    help="members per dynamics block in split mode")
     args = p.parse_args()
@@ -62,7 +63,7 @@ def main():
     rs = np.random.RandomState(0)
     eng = GpuEngine(cfg, comm, policy, nt, env, rs, use_graph=args.graph,
                     rollout_mode="episode" if args.episode else "step",
-                    split_dyn=args.split or None)
+                    split_dyn=args.split or None, pair_rollout=args.pair or None)
     ranker = CenteredRanker()
     eng.step(ranker)  # warmup
 
