@@ -83,7 +83,7 @@ def main():
         "policy": {"layer_sizes": list(args.layers), "ac_std": 0.01, "l2coeff": 0.005,
                    "lr": 0.01, "ob_clip": 5, "save_obs_chance": 0.01},
         "general": {"name": "bench", "policies_per_gen": ppg, "batch_size": 500,
-                    "seed": global_seed},
+                    "seed": global_seed, "pair_rollout": not args.no_pair},
     })
 
     B = 2 * (ppg // world // 2) + 1
@@ -104,9 +104,11 @@ def main():
                                   device=device)
 
     if use_cuda:
+        # pair_rollout arrives via cfg so the engine's grid-size gate can
+        # fall back to the fused per-member path for tiny populations
         engine = GpuEngine(cfg, comm, policy, nt, env, rs, objective=args.objective,
                            use_graph=not args.no_graph,
-                           pair_rollout=not args.no_pair)
+                           pair_rollout=False if args.no_pair else None)
         if args.objective == "nsr":
             # seeded starter archive on device (NSR-A semantics: novelty vs
             # the behaviour archive, grown per generation)

@@ -206,8 +206,16 @@ class GpuEngine:
         # (mlp_layers_pair) — per-step HBM weight traffic halves vs the
         # materialized per-member blobs. Effective weights become
         # bf16(theta) +- bf16(sigma*eps) (two bf16 roundings instead of one).
-        want_pair = bool(cfg.general.get("pair_rollout", False)) if pair_rollout \
-            is None else bool(pair_rollout)
+        # constructor True FORCES pair (tests, explicit intent); the config
+        # flag is auto-gated on grid size: pair halves the block count, so
+        # tiny populations would idle CUs (Hopper pop 256 -> 128 pair blocks
+        # measured 9.0M vs 12.4M env-steps/s fused) — require >= 2 pair
+        # blocks per CU before the halved HBM traffic can win
+        if pair_rollout is not None:
+            want_pair = bool(pair_rollout)
+        else:
+            want_pair = (bool(cfg.general.get("pair_rollout", False))
+                         and self.pairs * self.eps >= 512)
         self.pair_rollout = (want_pair and self.fused and not self.split_dyn
                              and self.rollout_mode == "step"
                              and self.steps_per_launch == 1 and self.pairs >= 1)
