@@ -209,13 +209,14 @@ class GpuEngine:
         # constructor True FORCES pair (tests, explicit intent); the config
         # flag is auto-gated on grid size: pair halves the block count, so
         # tiny populations would idle CUs (Hopper pop 256 -> 128 pair blocks
-        # measured 9.0M vs 12.4M env-steps/s fused) — require >= 2 pair
-        # blocks per CU before the halved HBM traffic can win
+        # measured 9.0M vs 12.4M env-steps/s fused); measured crossover is
+        # ~384 pair blocks (1.5/CU): pop 512 fused 58.0 vs pair 58.9 us/step,
+        # pop 768 73.3 vs 70.3, pop 1024 89.6 vs 75.2
         if pair_rollout is not None:
             want_pair = bool(pair_rollout)
         else:
             want_pair = (bool(cfg.general.get("pair_rollout", False))
-                         and self.pairs * self.eps >= 512)
+                         and self.pairs * self.eps >= 384)
         self.pair_rollout = (want_pair and self.fused and not self.split_dyn
                              and self.rollout_mode == "step"
                              and self.steps_per_launch == 1 and self.pairs >= 1)
