@@ -1067,3 +1067,52 @@ def test_pair_rollout_sigma_close_and_deterministic(dev):
     c = _pair_engine_fits(dev, std=0.02, pair=True, gens=1, max_steps=10)
     np.testing.assert_array_equal(b[0], c[0])
     assert torch.equal(b[1], c[1])
+
+
+def test_pair_rollout_variants_sigma0_bitwise(dev):
+    """Pair path across slot-mapping variants — eps>1 (episode-averaged
+    slots), goal-conditioned obs, odd S (scalar dynamics fallback) — each
+    bitwise-equal to the fused path at sigma=0."""
+    import numpy as np
+    from es_pytorch_amd.config import AttrDict
+    from es_pytorch_amd.core.engine import GpuEngine
+    from es_pytorch_amd.core.noisetable import NoiseTable
+    from es_pytorch_amd.core.policy import Policy
+    from es_pytorch_amd.envs import make_batched
+    from es_pytorch_amd.nn.nn import FeedForward
+    from es_pytorch_amd.nn.optimizers import Adam
+    from es_pytorch_amd.parallel.comm import Comm
+    from es_pytorch_amd.utils.rankers import CenteredRanker
+
+    for env_name, eps, kw in (("Humanoid-v2", 2, {}),
+                              ("HumanoidFlagrunBulletEnv-v0", 1,
+                               {"goal_conditioned": True}),
+                              ("Walker2d-v3", 1, {})):
+        out = {}
+        for pair in (False, True):
+            torch.manual_seed(77)
+            comm = Comm(dev)
+            cfg = AttrDict({"env": {"name": env_name, "max_steps": 15},
+                            "noise": {"tbl_size": 600_000, "std": 0.0},
+                            "policy": {"layer_sizes": [32], "ac_std": 0.01,
+                                       "l2coeff": 0.005, "lr": 0.01, "ob_clip": 5,
+                                       "save_obs_chance": 1.0},
+                            "general": {"policies_per_gen": 6, "batch_size": 500,
+                                        "seed": 2, "eps_per_policy": eps}})
+            B = 7 * eps
+            env = make_batched(env_name, B, dev, max_steps=15, **kw)
+            nn = FeedForward([32], torch.nn.Tanh(), env, 0.01, 5)
+            policy = Policy(nn, 0.0, Adam(len(Policy.get_flat(nn)), 0.01))
+            nt = NoiseTable.create_shared(comm, 600_000, len(policy), seed=3,
+                                          device=dev)
+            rs = np.random.RandomState(78)
+            eng = GpuEngine(cfg, comm, policy, nt, env, rs, use_graph=False,
+                            pair_rollout=pair)
+            assert eng.pair_rollout == pair, env_name
+            ranker = CenteredRanker()
+            eng.step(ranker)
+            torch.cuda.synchronize(dev)
+            out[pair] = (np.concatenate([ranker.fits_pos, ranker.fits_neg]).ravel(),
+                         eng.rew_total.cpu().clone())
+        np.testing.assert_array_equal(out[False][0], out[True][0], err_msg=env_name)
+        assert torch.equal(out[False][1], out[True][1]), env_name
