@@ -1116,3 +1116,62 @@ def test_pair_rollout_variants_sigma0_bitwise(dev):
                          eng.rew_total.cpu().clone())
         np.testing.assert_array_equal(out[False][0], out[True][0], err_msg=env_name)
         assert torch.equal(out[False][1], out[True][1]), env_name
+
+
+def test_pair_rollout_act_modes_sigma0_bitwise(dev):
+    """Pair path with K9 binned decode and both integrated-gaussian modes:
+    sigma=0 bitwise-equal to the fused path (the decode runs per slot in
+    both, so the action noise/argmax indices must line up exactly)."""
+    import numpy as np
+    from es_pytorch_amd.config import AttrDict
+    from es_pytorch_amd.core.engine import GpuEngine
+    from es_pytorch_amd.core.noisetable import NoiseTable
+    from es_pytorch_amd.core.policy import Policy
+    from es_pytorch_amd.envs import make_batched
+    from es_pytorch_amd.nn.nn import (FFBinned, FFIntegGausAction,
+                                      FFIntegGausActionMulti)
+    from es_pytorch_amd.nn.optimizers import Adam
+    from es_pytorch_amd.parallel.comm import Comm
+    from es_pytorch_amd.utils.rankers import CenteredRanker
+
+    from es_pytorch_amd.spaces import Box
+
+    def build_net(kind, env):
+        if kind == "binned":
+            return FFBinned([32], torch.nn.Tanh(), env, n_bins=5, ob_clip=5)
+
+        class _View:  # engine contract: output layer = adim+1 / 2*adim
+            observation_space = env.observation_space
+            action_space = Box(-1.0, 1.0, (env.ac_dim + 1 if kind == "ig"
+                                           else 2 * env.ac_dim,))
+
+        cls = FFIntegGausAction if kind == "ig" else FFIntegGausActionMulti
+        return cls([32], torch.nn.Tanh(), _View, ac_std=0.0, ob_clip=5)
+
+    for kind in ("binned", "ig", "igm"):
+        out = {}
+        for pair in (False, True):
+            torch.manual_seed(81)
+            comm = Comm(dev)
+            cfg = AttrDict({"env": {"name": "Hopper-v3", "max_steps": 15},
+                            "noise": {"tbl_size": 500_000, "std": 0.0},
+                            "policy": {"layer_sizes": [32], "ac_std": 0.0,
+                                       "l2coeff": 0.005, "lr": 0.01, "ob_clip": 5,
+                                       "save_obs_chance": 1.0},
+                            "general": {"policies_per_gen": 6, "batch_size": 500,
+                                        "seed": 1}})
+            env = make_batched("Hopper-v3", 7, dev, max_steps=15,
+                               terminate_on_fall=False)
+            nn = build_net(kind, env)
+            policy = Policy(nn, 0.0, Adam(len(Policy.get_flat(nn)), 0.01))
+            nt = NoiseTable.create_shared(comm, 500_000, len(policy), seed=4,
+                                          device=dev)
+            rs = np.random.RandomState(82)
+            eng = GpuEngine(cfg, comm, policy, nt, env, rs, use_graph=False,
+                            pair_rollout=pair)
+            assert eng.pair_rollout == pair
+            ranker = CenteredRanker()
+            eng.step(ranker)
+            torch.cuda.synchronize(dev)
+            out[pair] = np.concatenate([ranker.fits_pos, ranker.fits_neg]).ravel()
+        np.testing.assert_array_equal(out[False], out[True], err_msg=kind)
