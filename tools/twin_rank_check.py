@@ -21,7 +21,6 @@ import sys
 
 os.environ["LOCAL_RANK"] = "0"  # both ranks deliberately share cuda:0
 
-import numpy as np
 import torch
 import torch.distributed as dist
 
