@@ -176,7 +176,10 @@ class RunCheckpointer:
         else:
             ests = state.get("engine_states",
                              [None] * self.comm.size)[self.comm.rank]
-            for e, est in zip(_as_list(engine), ests or []):
+            engines = _as_list(engine)
+            if not ests:  # snapshot written without engines (CPU path)
+                ests = [None] * len(engines)
+            for e, est in zip(engines, ests):
                 e.restore_from_policy(gen=state["next_gen"])
                 if est is not None:
                     e.load_checkpoint_state(est)

@@ -69,6 +69,7 @@ class Watchdog:
         with self._lock:
             self._label = label
             self._deadline = time.monotonic() + self.timeout_s
+            self._fired = False  # a custom on_timeout may not exit; re-arm
 
     def disarm(self):
         with self._lock:
