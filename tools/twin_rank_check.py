@@ -57,7 +57,10 @@ def main():
     nn = FeedForward([64], torch.nn.Tanh(), env, 0.01, 5)
     policy = Policy(nn, 0.02, Adam(len(Policy.get_flat(nn)), 0.01))
     nt = NoiseTable.create_shared(comm, 4_000_000, len(policy), seed=12, device=dev)
-    eng = GpuEngine(cfg, comm, policy, nt, env, rs, use_graph=True)
+    import sys as _sys
+    pair = "--pair" in _sys.argv
+    eng = GpuEngine(cfg, comm, policy, nt, env, rs, use_graph=True,
+                    pair_rollout=True if pair else None)
     ranker = CenteredRanker()
 
     for _ in range(3):
@@ -74,8 +77,9 @@ def main():
     assert offs[0] != offs[1], "ranks drew identical noise offsets"
 
     if comm.rank == 0:
-        print(f"TWIN-RANK OK: 3 gens, params sha256={digest[:16]}, "
-              f"moments sha256={opt_digest[:16]}, identical on both ranks")
+        print(f"TWIN-RANK OK (pair={eng.pair_rollout}): 3 gens, "
+              f"params sha256={digest[:16]}, moments sha256={opt_digest[:16]}, "
+              "identical on both ranks")
     dist.destroy_process_group()
     return 0
 
