@@ -34,3 +34,29 @@ def test_merge_override():
 def test_to_dict_roundtrip():
     d = {"a": {"b": [1, 2]}, "c": "x"}
     assert AttrDict(d).to_dict() == d
+
+
+def test_every_shipped_config_is_well_formed():
+    """Every configs/*.json loads, names a known env, and carries the keys
+    the entry scripts read — catches config drift as options evolve."""
+    import os
+
+    from es_pytorch_amd.envs import make_batched
+
+    root = os.path.join(os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
+                        "configs")
+    files = sorted(f for f in os.listdir(root) if f.endswith(".json"))
+    assert len(files) >= 15
+    for f in files:
+        cfg = load_config(os.path.join(root, f))
+        if f == "batch.json":  # sweep driver schema, not a run config
+            continue
+        if f == "multi_agent.json":  # PursuitTag lives in envs.multiagent
+            from es_pytorch_amd.envs.multiagent import BatchedPursuitTag
+            assert BatchedPursuitTag(3, "cpu").N_AGENTS >= 2
+            continue
+        assert cfg.general.policies_per_gen % 2 == 0, f
+        assert cfg.noise.tbl_size > 0 and 0 < cfg.noise.std < 1, f
+        assert cfg.policy.layer_sizes and cfg.policy.lr > 0, f
+        env = make_batched(cfg.env.name, 3, "cpu", max_steps=5)
+        assert env.ob_dim > 0, f
