@@ -4,7 +4,13 @@ import pytest
 import torch
 
 fastapi = pytest.importorskip("fastapi")
-from fastapi.testclient import TestClient  # noqa: E402
+import warnings  # noqa: E402
+
+with warnings.catch_warnings():
+    # this image's fastapi/starlette pairing warns about its own TestClient
+    # import; the deprecation is theirs, not ours
+    warnings.simplefilter("ignore")
+    from fastapi.testclient import TestClient  # noqa: E402
 
 from es_pytorch_amd.core.policy import Policy  # noqa: E402
 from es_pytorch_amd.envs import make  # noqa: E402
