@@ -1,10 +1,15 @@
 """Polymorphic "what is fitness" result types.
 
-Same hierarchy as the reference (``src/gym/training_result.py:9-97``): a
-rollout produces rewards, a behaviour trace (3-D positions per step), saved
-observations and a step count; subclasses define the fitness objective(s) —
-total reward, mean reward/step, distance, x-displacement, novelty, or the
-2-objective [reward, novelty] used by NSR/NSRA.
+The class names, constructor signature and property surface match the
+reference hierarchy (``src/gym/training_result.py:9-97``) because the entry
+scripts and ``es.step`` are generic over them; the bodies are this repo's own.
+
+A rollout yields four raw artifacts — per-step rewards, a flat behaviour
+trace of 3-D positions ``[x0,y0,z0, x1,y1,z1, ...]``, the observations kept
+for normalization statistics, and the step count. Each subclass reduces those
+to its fitness objective(s): total reward, per-step reward, planar distance,
+x-displacement, novelty against an archive, or the 2-objective
+[reward, novelty] pair that NSR/NSRA rank jointly.
 """
 from __future__ import annotations
 
@@ -17,88 +22,119 @@ from es_pytorch_amd.utils.novelty import novelty
 
 
 class TrainingResult(ABC):
-    """Result of a single policy evaluation (reference ``training_result.py:9-29``)."""
+    """Raw artifacts of one policy evaluation + the fitness reduction."""
 
     def __init__(self, rewards: List[float], positions: List[float], obs: np.ndarray,
                  steps: int, *args, **kwargs):
-        self.rewards: List[float] = rewards
-        self.positions: List[float] = positions  # flat [x0,y0,z0, x1,y1,z1, ...]
-        self.obs: np.ndarray = obs
+        self.rewards = rewards
+        self.positions = positions  # flat [x0,y0,z0, x1,y1,z1, ...]
+        self.obs = obs
         self.steps = steps
 
     @property
     def ob_sum_sq_cnt(self) -> Tuple[np.ndarray, np.ndarray, int]:
-        cnt = len(self.obs) if np.any(self.obs) else 0
-        return self.obs.sum(axis=0), np.square(self.obs).sum(axis=0), cnt
+        """(Σob, Σob², count) feeding the ObStat merge. An all-zero obs block
+        means "nothing was saved this episode" (save_obs_chance miss) and
+        must contribute count 0, not len(obs) zeros."""
+        ob = np.asarray(self.obs)
+        n = 0 if not ob.any() else ob.shape[0]
+        return ob.sum(axis=0), (ob * ob).sum(axis=0), n
 
     @abstractmethod
     def get_result(self) -> List[float]:
-        ...
+        """The fitness objective vector this evaluation scored."""
 
-    result: List[float] = property(lambda self: self.get_result())
-    reward = property(lambda self: sum(self.rewards))
-    behaviour = property(lambda self: self.positions[-3:-1])  # final (x, y)
+    @property
+    def result(self) -> List[float]:
+        return self.get_result()
+
+    @property
+    def reward(self):
+        return sum(self.rewards)
+
+    @property
+    def behaviour(self):
+        """Final planar (x, y) — the novelty-search behaviour descriptor."""
+        return self.positions[-3:-1]
 
 
 class MultiAgentTrainingResult(TrainingResult):
-    """Joint result of a co-evolution rollout (reference ``training_result.py:32-59``)."""
+    """Joint result of a co-evolution rollout: ``rewards`` and ``obs`` carry
+    a trailing per-agent axis; reductions happen per column."""
 
     def get_result(self):
         return self.reward
 
     @property
+    def reward(self):
+        # one total-reward entry per agent
+        return np.asarray(self.rewards).sum(axis=0).tolist()
+
+    @property
     def ob_sum_sq_cnt(self):
-        out = []
-        for i in range(self.obs.shape[1]):
-            curr = self.obs[:, i]
-            cnt = len(curr) if np.any(curr) else 0
-            out.append((curr.sum(axis=0), np.square(curr).sum(axis=0), cnt))
-        return out
+        ob = np.asarray(self.obs)
+        return [self._agent_stats(ob[:, a]) for a in range(ob.shape[1])]
+
+    @staticmethod
+    def _agent_stats(ob: np.ndarray) -> Tuple[np.ndarray, np.ndarray, int]:
+        n = 0 if not ob.any() else ob.shape[0]
+        return ob.sum(axis=0), (ob * ob).sum(axis=0), n
 
     def trainingresults(self, tr_type: Type[TrainingResult]) -> List[TrainingResult]:
-        rews, obs = np.array(self.rewards), np.array(self.obs)
-        return [tr_type(rews[:, i], self.positions, obs[:, i], self.steps)
-                for i in range(np.array(self.rewards).shape[1])]
-
-    reward = property(lambda self: np.sum(self.rewards, axis=0).tolist())
+        """Split into one single-agent result per agent (shared behaviour/steps)."""
+        rews = np.asarray(self.rewards)
+        obs = np.asarray(self.obs)
+        return [tr_type(rews[:, a], self.positions, obs[:, a], self.steps)
+                for a in range(rews.shape[1])]
 
 
 class RewardResult(TrainingResult):
+    """Fitness = total episode reward."""
+
     def get_result(self) -> List[float]:
         return [self.reward]
 
 
 class MeanRewardResult(TrainingResult):
+    """Fitness = reward per step survived (discourages early termination)."""
+
     def get_result(self) -> List[float]:
         return [self.reward / self.steps]
 
 
 class DistResult(TrainingResult):
+    """Fitness = planar distance of the final position from the origin."""
+
     def get_result(self) -> List[float]:
-        return [float(np.linalg.norm(self.positions[-3:-1]))]
+        x, y = self.positions[-3], self.positions[-2]
+        return [float(np.hypot(x, y))]
 
 
 class XDistResult(DistResult):
+    """Fitness = signed x-displacement (directed locomotion)."""
+
     def get_result(self) -> List[float]:
         return [self.positions[-3]]
 
 
 class NSResult(TrainingResult):
-    """Novelty-only fitness (reference ``training_result.py:82-92``)."""
+    """Fitness = novelty of the behaviour descriptor vs the archive."""
 
     def __init__(self, rewards, positions, obs, steps, archive: np.ndarray, k: int):
         super().__init__(rewards, positions, obs, steps)
         self.archive = archive
         self.k = k
 
-    novelty = property(lambda self: novelty(np.array(self.behaviour), self.archive, self.k))
+    @property
+    def novelty(self):
+        return novelty(np.asarray(self.behaviour), self.archive, self.k)
 
     def get_result(self) -> List[float]:
         return [self.novelty]
 
 
 class NSRResult(NSResult):
-    """[reward, novelty] 2-objective fitness (reference ``training_result.py:95-97``)."""
+    """2-objective fitness [reward, novelty]; NSR/NSRA weight the two."""
 
     def get_result(self) -> List[float]:
-        return [sum(self.rewards), self.novelty]
+        return [self.reward, self.novelty]
