@@ -30,6 +30,23 @@ def init_normal(m):
         torch.nn.init.kaiming_normal_(m.weight)
 
 
+#: reference module path -> this package (attribute-compatible classes)
+_REF_MODULE_MAP = {
+    "src.core.policy": "es_pytorch_amd.core.policy",
+    "src.nn.nn": "es_pytorch_amd.nn.nn",
+    "src.nn.obstat": "es_pytorch_amd.nn.obstat",
+    "src.nn.optimizers": "es_pytorch_amd.nn.optimizers",
+    "src.gym.training_result": "es_pytorch_amd.rollout.results",
+}
+
+
+class _CompatUnpickler(pickle.Unpickler):
+    """Unpickler that accepts the reference's module paths (cross-load)."""
+
+    def find_class(self, module: str, name: str):
+        return super().find_class(_REF_MODULE_MAP.get(module, module), name)
+
+
 class Policy:
     def __init__(self, module: BaseNet, noise_std: float, optim: Optimizer):
         module.apply(init_normal)
@@ -50,8 +67,17 @@ class Policy:
 
     @staticmethod
     def load(file: str) -> "Policy":
+        """Open a checkpoint written by this repo OR by the reference.
+
+        A pickle produced by the reference's ``Policy.save``
+        (``src/core/policy.py:43-47``) stores classes under the reference's
+        module paths (``src.core.policy.Policy``, ``src.nn.nn.FeedForward``,
+        ...); :class:`_CompatUnpickler` remaps those onto this package's
+        attribute-compatible classes, so reference checkpoints load
+        byte-for-byte with no conversion step.
+        """
         with open(file, "rb") as f:
-            policy: Policy = pickle.load(f)
+            policy: Policy = _CompatUnpickler(f).load()
         policy.set_nn_params(policy.flat_params)
         return policy
 
