@@ -114,6 +114,19 @@ class GpuEngine:
         assert env.batch == self.B, f"env batch {env.batch} != engine batch {self.B}"
 
         self.dims = policy._module.layer_dims()
+        # config-time surface of the HIP kernels' compile-time shape limits
+        # (mlp_core.h ES_MAXL/ES_MAXDIM) — fail before any launch, not with a
+        # kernel error code mid-generation
+        if len(self.dims) - 1 > 8:
+            raise ValueError(
+                f"model has {len(self.dims) - 1} layers; the fused HIP forward "
+                f"supports at most 8 (ES_MAXL in ops/csrc/hip/mlp_core.h) — "
+                f"reduce policy.layer_sizes or raise ES_MAXL and rebuild")
+        if max(self.dims) > 2048:
+            raise ValueError(
+                f"widest layer is {max(self.dims)}; the fused HIP forward "
+                f"supports at most 2048 (ES_MAXDIM in ops/csrc/hip/mlp_core.h) "
+                f"— narrow policy.layer_sizes or raise ES_MAXDIM and rebuild")
         # K9: FFBinned policies emit adim*bins logits decoded in-kernel
         self.bins = int(getattr(policy._module, "bins", 0))
         # integrated-gaussian-action variants (reference nn.py:53-96): the net
@@ -223,15 +236,16 @@ class GpuEngine:
         self.pair_rollout = (want_pair and self.fused and not self.split_dyn
                              and self.rollout_mode == "step"
                              and self.steps_per_launch == 1 and self.pairs >= 1)
+        self._zero_off = torch.zeros(1, dtype=torch.int64, device=d)
+        self._zero_sign = torch.zeros(1, dtype=torch.float32, device=d)
         if self.pair_rollout:
             self.theta_row = torch.empty((1, self.row_stride), dtype=torch.bfloat16,
                                          device=d)
             self.eps_rows = torch.empty((self.pairs, self.row_stride),
                                         dtype=torch.bfloat16, device=d)
             self._zeros_n = torch.zeros(self.n, dtype=torch.float32, device=d)
-            self._zero_off = torch.zeros(1, dtype=torch.int64, device=d)
-            self._zero_sign = torch.zeros(1, dtype=torch.float32, device=d)
             self._one_signs = torch.ones(self.pairs, dtype=torch.float32, device=d)
+        self._warned_host_ranker = False
 
     # ------------------------------------------------------------------ ops
     def _stream(self):
@@ -483,6 +497,47 @@ class GpuEngine:
             return torch.stack([rew, nov], dim=1)
         raise ValueError(f"unknown objective {self.objective!r}")
 
+    def noiseless_eval(self) -> Tuple[float, np.ndarray, int]:
+        """Evaluation-only noiseless episode: (reward, behaviour(3,), steps).
+
+        Runs ONLY the unperturbed policy — no optimizer update, no numpy RNG
+        draws, nothing mutated but the rollout scratch buffers. Mirrors the
+        reference's archive-init evaluation (``nsra.py:31-45``: noiseless
+        rollouts with no training step).
+        """
+        if self.pair_rollout:
+            ops.check(ops.hip().es_pheno_bf16(
+                self.theta_row.data_ptr(), self.theta.data_ptr(),
+                self.nt.noise.data_ptr(), self._zero_off.data_ptr(),
+                self._zero_sign.data_ptr(), 1, self.n, self.row_stride, 0.0,
+                self._stream()), "es_pheno_bf16")
+        elif self.device.type == "cuda":
+            # bf16(theta) into the noiseless slot's weights row (sign 0)
+            ops.check(ops.hip().es_pheno_bf16(
+                self.weights[self.M - 1:].data_ptr(), self.theta.data_ptr(),
+                self.nt.noise.data_ptr(), self._zero_off.data_ptr(),
+                self._zero_sign.data_ptr(), 1, self.n, self.row_stride, 0.0,
+                self._stream()), "es_pheno_bf16")
+        else:
+            self.weights[self.M - 1, :self.n] = self.theta.bfloat16()
+        self._reset_rollout_state()
+        self.obs_buf.copy_(self.env.reset(self._gen_seed()))
+        self.acstd_dev.fill_(0.0)  # reference noiseless eval: use_ac_noise False
+        if self.fused:
+            self._loco_noiseless_episode()
+        else:
+            # generic torch path computes the whole batch; only the noiseless
+            # slots are read out (init-time only, cost irrelevant)
+            self.weights.copy_(self.weights[self.M - 1:].expand_as(self.weights))
+            for t in range(self.max_steps):
+                self._step_body(t)
+        if self.device.type == "cuda":
+            torch.cuda.synchronize(self.device)
+        rew = float(self._member_rewards()[-1].item())
+        behv = self._member_behv()[-1].cpu().numpy()
+        steps = int(self.member_steps[(self.M - 1) * self.eps:].sum().item())
+        return rew, behv, steps
+
     # ---------------------------------------------------------------- step
     def step(self, ranker: Ranker, reporter=None) -> Tuple[_NoiselessResult, ObStat]:
         """Run one full generation; mirrors ``es.step`` semantics (``es.py:23-51``)."""
@@ -561,6 +616,14 @@ class GpuEngine:
             ranker.ranked_fits = rf.cpu().numpy().astype(np.float64)
             ranker.n_fits_ranked = npop
         else:
+            if not self._warned_host_ranker and self.device.type == "cuda":
+                self._warned_host_ranker = True
+                if type(ranker) is not CenteredRanker and self.comm.rank == 0:
+                    import sys
+                    print(f"[engine] ranking on host: {type(ranker).__name__} "
+                          f"(device fast path covers plain single-objective "
+                          f"CenteredRanker only; pop-sized, sub-ms)",
+                          file=sys.stderr)
             all_rows = all_rows_dev.cpu().numpy()
             pos, neg = all_rows[:, :O], all_rows[:, O:2 * O]
             inds = all_rows[:, -1]

@@ -46,7 +46,22 @@ class Optimizer(ABC):
 
 
 class SimpleES(Optimizer):
-    """Plain scaled-gradient step (reference ``optimizers.py:28-33``)."""
+    """Plain scaled-gradient step (reference ``optimizers.py:28-33``).
+
+    WARNING (sign convention): returns ``+lr*g`` while SGD/Adam return the
+    NEGATIVE step — swapping SimpleES into a caller that passes
+    ``l2coeff*theta - grad`` (``es.approx_grad``) would DESCEND on fitness.
+    A loud warning is emitted at construction; pass a pre-negated input or
+    use SGD/Adam for the training path (GpuEngine rejects SimpleES)."""
+
+    def __init__(self, dim: int, lr: float):
+        super().__init__(dim, lr)
+        import warnings
+        warnings.warn(
+            "SimpleES returns +lr*g (opposite sign convention to SGD/Adam); "
+            "with the standard es.approx_grad input it descends on fitness — "
+            "negate the input or use SGD/Adam for training",
+            UserWarning, stacklevel=2)
 
     def _compute_step(self, globalg: np.ndarray) -> np.ndarray:
         return self.lr * globalg

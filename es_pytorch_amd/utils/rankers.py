@@ -19,10 +19,17 @@ import numpy as np
 
 
 def rank(x: np.ndarray) -> np.ndarray:
-    """Ranks in [0, len(x)) (reference ``rankers.py:9-17``)."""
+    """Ranks in [0, len(x)) (reference ``rankers.py:9-17``).
+
+    STABLE sort so tied fitnesses (common early, when many members terminate
+    with identical reward) rank identically to the engine's device fast path
+    (torch.argsort(stable=True), ``core/engine.py``) — GPU-vs-host runs stay
+    comparable. The reference uses numpy's default introsort, which permutes
+    ties arbitrarily; rank VALUES are identical either way.
+    """
     assert x.ndim == 1
     ranks = np.empty(len(x), dtype=int)
-    ranks[x.argsort()] = np.arange(len(x))
+    ranks[x.argsort(kind="stable")] = np.arange(len(x))
     return ranks
 
 

@@ -115,12 +115,11 @@ def main(cfg):
     if ck_state is not None:
         pass  # resumed runs restore the archive instead of re-initializing
     elif use_gpu:
-        # one cheap generation per policy yields its noiseless behaviour
-        ranker0 = MultiObjectiveRanker(CenteredRanker(), cfg.nsr.initial_w)
-        for i, eng in enumerate(engines):
-            eng.archive = torch.zeros((1, 2), dtype=torch.float64, device=device)
-            tr, _ = eng.step(ranker0)
-            b = comm.broadcast_obj(list(tr.behaviour), src=0)
+        # evaluation-only noiseless episode per policy (reference
+        # init_archive, nsra.py:31-45: no training step, no RNG consumption)
+        for eng in engines:
+            _, b3, _ = eng.noiseless_eval()
+            b = comm.broadcast_obj([float(b3[0]), float(b3[1])], src=0)
             archive = update_archive(None, b, archive)
             policies_novelties.append(max(1e-2, novelty(np.array(b), archive,
                                                         cfg.novelty.k)))

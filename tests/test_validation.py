@@ -1,0 +1,60 @@
+"""Config-time guards and tie-handling determinism (round-2 hardening)."""
+import numpy as np
+import pytest
+import torch
+
+from es_pytorch_amd.config import AttrDict
+from es_pytorch_amd.core.engine import GpuEngine
+from es_pytorch_amd.core.noisetable import NoiseTable
+from es_pytorch_amd.core.policy import Policy
+from es_pytorch_amd.envs import make_batched
+from es_pytorch_amd.nn.nn import FeedForward
+from es_pytorch_amd.nn.optimizers import Adam, SimpleES
+from es_pytorch_amd.parallel.comm import Comm
+from es_pytorch_amd.utils.rankers import rank
+
+
+def _setup(layer_sizes):
+    comm = Comm(torch.device("cpu"))
+    cfg = AttrDict({
+        "env": {"name": "Hopper-v2", "max_steps": 8},
+        "noise": {"tbl_size": 10000, "std": 0.02},
+        "policy": {"layer_sizes": layer_sizes, "ac_std": 0.0, "l2coeff": 0.005,
+                   "lr": 0.01, "ob_clip": 5},
+        "general": {"name": "t", "policies_per_gen": 4, "batch_size": 100},
+    })
+    env = make_batched("Hopper-v2", 5, torch.device("cpu"), max_steps=8)
+    nn = FeedForward(layer_sizes, torch.nn.Tanh(), env, 0.0, 5)
+    policy = Policy(nn, 0.02, Adam(len(Policy.get_flat(nn)), 0.01))
+    nt = NoiseTable.create_shared(comm, 10000, len(policy), seed=1)
+    rs = np.random.RandomState(0)
+    return cfg, comm, policy, nt, env, rs
+
+
+def test_too_many_layers_rejected_at_construction():
+    with pytest.raises(ValueError, match="at most 8"):
+        GpuEngine(*_setup([8] * 9), use_graph=False)
+
+
+def test_too_wide_layer_rejected_at_construction():
+    with pytest.raises(ValueError, match="at most 2048"):
+        GpuEngine(*_setup([4096]), use_graph=False)
+
+
+def test_max_shape_accepted():
+    GpuEngine(*_setup([16] * 7), use_graph=False)  # 7 hidden + in/out = 8 layers: OK
+
+
+def test_simple_es_warns_about_sign_convention():
+    with pytest.warns(UserWarning, match="sign convention|descends"):
+        SimpleES(10, 0.1)
+
+
+def test_rank_breaks_ties_stably():
+    """Tied fitnesses must rank identically to torch.argsort(stable=True),
+    the engine's device fast path — first occurrence gets the lower rank."""
+    x = np.array([1.0, 0.5, 1.0, 0.5, 2.0])
+    r = rank(x)
+    t = torch.empty(5, dtype=torch.long)
+    t[torch.argsort(torch.from_numpy(x), stable=True)] = torch.arange(5)
+    np.testing.assert_array_equal(r, t.numpy())
