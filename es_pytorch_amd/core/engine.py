@@ -139,6 +139,16 @@ class GpuEngine:
             self.act_mode = 2
         else:
             self.act_mode = 0
+        # single output-layer contract across engine and episodic paths
+        # (nn.py _ActionView): verify the module was sized for THIS env
+        out = self.dims[-1]
+        expect = {0: env.ac_dim if self.bins <= 1 else env.ac_dim * self.bins,
+                  2: env.ac_dim + 1, 3: 2 * env.ac_dim}[self.act_mode]
+        if out != expect:
+            raise ValueError(
+                f"output layer is {out} wide but act_mode {self.act_mode} on a "
+                f"{env.ac_dim}-action env needs {expect} — was the module built "
+                f"for a different env?")
         self.n = int(np.sum([I * O + O for I, O in zip(self.dims[:-1], self.dims[1:])]))
         assert self.n == len(policy), (self.n, len(policy))
         self.perm = forward_perm(self.dims).to(self.device)  # fwd idx -> flat idx

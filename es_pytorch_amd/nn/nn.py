@@ -82,8 +82,34 @@ class FeedForward(BaseNet):
         return a
 
 
+class _ActionView:
+    """Env facade that widens the action space seen by FeedForward sizing.
+
+    The integrated-gaussian variants consume part of the network output as
+    the action std, so their output layer must be WIDER than the env's
+    action dim: adim+1 (one shared std) or 2*adim (per-dim stds). The
+    reference sized the output layer adim (``nn.py:33`` reused by ``:53``)
+    and then consumed one output as std, emitting adim-1 actions — an
+    off-by-one no real env accepts. Here BOTH the episodic and the engine
+    path size via this view and emit exactly adim actions (single
+    output-layer contract; the engine's act decode assumes it).
+    """
+
+    def __init__(self, env, extra: int):
+        self.observation_space = env.observation_space
+        space = env.action_space
+        shape = (int(np.prod(space.shape)) + extra,)
+        self.action_space = type("_W", (), {"shape": shape})()
+
+
 class FFIntegGausAction(FeedForward):
-    """MLP whose FIRST output is the (shared) action std (reference ``nn.py:53-74``)."""
+    """MLP whose FIRST output is the (shared) action std; remaining outputs
+    are the adim action means (reference ``nn.py:53-74``; output sizing
+    unified across paths, see :class:`_ActionView`)."""
+
+    def __init__(self, layer_sizes: List[int], activation: nn.Module, env, ac_std: float,
+                 ob_clip: float = 5):
+        super().__init__(layer_sizes, activation, _ActionView(env, 1), ac_std, ob_clip)
 
     def forward(self, inp: Tensor, **kwargs) -> np.ndarray:
         rs: Optional[np.random.RandomState] = kwargs.get("rs")
@@ -95,7 +121,13 @@ class FFIntegGausAction(FeedForward):
 
 
 class FFIntegGausActionMulti(FeedForward):
-    """MLP emitting [mean-half, std-half] outputs (reference ``nn.py:77-96``)."""
+    """MLP emitting [mean-half, std-half] outputs (reference ``nn.py:77-96``;
+    output layer sized 2*adim on both paths, see :class:`_ActionView`)."""
+
+    def __init__(self, layer_sizes: List[int], activation: nn.Module, env, ac_std: float,
+                 ob_clip: float = 5):
+        adim = int(np.prod(env.action_space.shape))
+        super().__init__(layer_sizes, activation, _ActionView(env, adim), ac_std, ob_clip)
 
     def forward(self, inp: Tensor, **kwargs) -> np.ndarray:
         rs: Optional[np.random.RandomState] = kwargs.get("rs")

@@ -518,11 +518,8 @@ def test_integ_gauss_engine(dev):
                         "general": {"policies_per_gen": 8, "batch_size": 100, "seed": 2}})
         env = make_batched("Hopper-v3", 9, dev, max_steps=25, terminate_on_fall=False)
 
-        class _View:  # engine contract: output layer = adim + 1
-            observation_space = env.observation_space
-            action_space = Box(-1.0, 1.0, (env.ac_dim + 1,))
-
-        nn = FFIntegGausAction([32], torch.nn.Tanh(), _View, ac_std=0.0, ob_clip=5)
+        # unified output contract: the net sizes its own output adim+1
+        nn = FFIntegGausAction([32], torch.nn.Tanh(), env, ac_std=0.0, ob_clip=5)
         policy = Policy(nn, 0.05, Adam(len(Policy.get_flat(nn)), 0.01))
         nt = NoiseTable.create_shared(comm, 400_000, len(policy), seed=8, device=dev)
         rs = np.random.RandomState(23)
@@ -1140,13 +1137,9 @@ def test_pair_rollout_act_modes_sigma0_bitwise(dev):
         if kind == "binned":
             return FFBinned([32], torch.nn.Tanh(), env, n_bins=5, ob_clip=5)
 
-        class _View:  # engine contract: output layer = adim+1 / 2*adim
-            observation_space = env.observation_space
-            action_space = Box(-1.0, 1.0, (env.ac_dim + 1 if kind == "ig"
-                                           else 2 * env.ac_dim,))
-
+        # unified output contract: these nets size their own output layer
         cls = FFIntegGausAction if kind == "ig" else FFIntegGausActionMulti
-        return cls([32], torch.nn.Tanh(), _View, ac_std=0.0, ob_clip=5)
+        return cls([32], torch.nn.Tanh(), env, ac_std=0.0, ob_clip=5)
 
     for kind in ("binned", "ig", "igm"):
         out = {}

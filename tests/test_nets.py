@@ -43,8 +43,11 @@ def test_integ_gaus_action():
     nn = FFIntegGausAction([8], torch.nn.Tanh(), _Env, ac_std=0.0, ob_clip=5)
     ob = torch.zeros(6)
     a = nn(ob, rs=np.random.RandomState(0))
-    # first output consumed as std -> adim-1 actions (reference nn.py:70-71)
-    assert a.shape == (2,)
+    # unified contract: output layer sized adim+1 (first output = std),
+    # so the env's full adim actions come out — the reference's sizing
+    # (nn.py:33 reused by :53) emitted adim-1, which no env accepts
+    assert nn.layer_dims()[-1] == 4
+    assert a.shape == (3,)
 
 
 def test_integ_gaus_action_multi():
@@ -56,7 +59,18 @@ def test_integ_gaus_action_multi():
 
     nn = FFIntegGausActionMulti([8], torch.nn.Tanh(), _Env4, ac_std=0.0, ob_clip=5)
     a = nn(torch.zeros(6), rs=np.random.RandomState(0))
-    assert a.shape == (2,)  # half mean, half std (reference nn.py:90-91)
+    assert nn.layer_dims()[-1] == 8  # [mean-half | std-half]
+    assert a.shape == (4,)
+
+
+def test_integ_gaus_sizing_matches_engine_contract():
+    """Cross-path parity: the SAME env must produce the SAME network on the
+    episodic path and the engine path (engine act decode modes 2/3 expect
+    adim+1 / 2*adim outputs; core/engine.py output-contract check)."""
+    for cls, extra in ((FFIntegGausAction, 1), (FFIntegGausActionMulti, 3)):
+        nn = cls([8], torch.nn.Tanh(), _Env, ac_std=0.0, ob_clip=5)
+        dims = nn.layer_dims()
+        assert dims == [6, 8, 3 + extra]
 
 
 def test_binned_decode():
