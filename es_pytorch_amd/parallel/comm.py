@@ -45,7 +45,10 @@ class Comm:
             self.backend = None
         if device is None:
             if torch.cuda.is_available():
-                local = int(os.environ.get("LOCAL_RANK", self.rank % max(1, torch.cuda.device_count())))
+                # modulo device count: N ranks on an M<N-GPU box (twin-rank
+                # integration tests) share devices instead of crashing
+                local = int(os.environ.get("LOCAL_RANK",
+                                           self.rank)) % max(1, torch.cuda.device_count())
                 device = torch.device("cuda", local)
             else:
                 device = torch.device("cpu")
@@ -111,10 +114,14 @@ def init_comm(device: Optional[torch.device] = None, timeout_s: int = 600) -> Co
     """
     ws = int(os.environ.get("WORLD_SIZE", "1"))
     if ws > 1 and not (dist.is_available() and dist.is_initialized()):
-        backend = "nccl" if torch.cuda.is_available() else "gloo"
+        # ES_COMM_BACKEND=gloo forces gloo transport while compute stays on
+        # the GPU — lets N ranks share ONE physical GPU for integration tests
+        # (RCCL, like NCCL, rejects two ranks on the same device)
+        backend = os.environ.get(
+            "ES_COMM_BACKEND", "nccl" if torch.cuda.is_available() else "gloo")
         if torch.cuda.is_available():
             local = int(os.environ.get("LOCAL_RANK", "0"))
-            torch.cuda.set_device(local)
+            torch.cuda.set_device(local % max(1, torch.cuda.device_count()))
         dist.init_process_group(backend=backend, timeout=datetime.timedelta(seconds=timeout_s))
     return Comm(device)
 

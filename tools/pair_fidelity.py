@@ -16,7 +16,7 @@ import torch
 sys.path.insert(0, __file__.rsplit("/", 2)[0])
 
 
-def run(horizon):
+def run(horizon, std=0.02, tbl=250_000_000):
     from es_pytorch_amd.config import AttrDict
     from es_pytorch_amd.core.engine import GpuEngine
     from es_pytorch_amd.core.noisetable import NoiseTable
@@ -32,7 +32,7 @@ def run(horizon):
         torch.manual_seed(99)
         comm = Comm(torch.device("cuda", 0))
         cfg = AttrDict({"env": {"name": "Humanoid-v2", "max_steps": horizon},
-                        "noise": {"tbl_size": 250_000_000, "std": 0.02},
+                        "noise": {"tbl_size": tbl, "std": std},
                         "policy": {"layer_sizes": [256, 256], "ac_std": 0.01,
                                    "l2coeff": 0.005, "lr": 0.01, "ob_clip": 5,
                                    "save_obs_chance": 0.01},
@@ -41,8 +41,8 @@ def run(horizon):
         env = make_batched("Humanoid-v2", 1281, comm.device, max_steps=horizon,
                            terminate_on_fall=True)
         nn = FeedForward([256, 256], torch.nn.Tanh(), env, 0.01, 5)
-        policy = Policy(nn, 0.02, Adam(len(Policy.get_flat(nn)), 0.01))
-        nt = NoiseTable.create_shared(comm, 250_000_000, len(policy), seed=9,
+        policy = Policy(nn, std, Adam(len(Policy.get_flat(nn)), 0.01))
+        nt = NoiseTable.create_shared(comm, tbl, len(policy), seed=9,
                                       device=comm.device)
         rs = np.random.RandomState(100)
         eng = GpuEngine(cfg, comm, policy, nt, env, rs, use_graph=False,
@@ -56,9 +56,15 @@ def run(horizon):
     rho = spearmanr(out[False][0], out[True][0]).correlation
     g0, g1 = out[False][1], out[True][1]
     cos = float(np.dot(g0, g1) / (np.linalg.norm(g0) * np.linalg.norm(g1)))
-    print(f"horizon {horizon:5d}: fitness spearman {rho:.4f}  grad cosine {cos:.4f}")
+    print(f"sigma {std:.3f} horizon {horizon:5d}: fitness spearman {rho:.4f}  "
+          f"grad cosine {cos:.4f}", flush=True)
+    return rho, cos
 
 
 if __name__ == "__main__":
+    # horizon sweep at the flagship sigma + sigma sweep at the flagship
+    # horizon (one operating point is not evidence)
     for h in (100, 1000):
-        run(h)
+        run(h, std=0.02, tbl=50_000_000)
+    for s in (0.005, 0.05, 0.1):
+        run(1000, std=s, tbl=50_000_000)
