@@ -49,6 +49,12 @@ def main():
                         "measured 1.27x over per-member weight blobs, and "
                         "the perturbation keeps MORE bf16 mantissa)")
     p.add_argument("--cpu", action="store_true", help="debug: run the engine on CPU")
+    p.add_argument("--mode", type=str, default="step", choices=["step", "episode"],
+                   help="rollout launch shape: per-step graph-replayed grid, or "
+                        "one whole-episode launch per generation (blocks drift; "
+                        "phases overlap chip-wide)")
+    p.add_argument("--pair-block", type=int, default=256, choices=[128, 256],
+                   help="pair-episode kernel block size")
     args = p.parse_args()
 
     from es_pytorch_amd.config import AttrDict
@@ -83,7 +89,8 @@ def main():
         "policy": {"layer_sizes": list(args.layers), "ac_std": 0.01, "l2coeff": 0.005,
                    "lr": 0.01, "ob_clip": 5, "save_obs_chance": 0.01},
         "general": {"name": "bench", "policies_per_gen": ppg, "batch_size": 500,
-                    "seed": global_seed, "pair_rollout": not args.no_pair},
+                    "seed": global_seed, "pair_rollout": not args.no_pair,
+                    "pair_block": args.pair_block},
     })
 
     B = 2 * (ppg // world // 2) + 1
@@ -107,7 +114,7 @@ def main():
         # pair_rollout arrives via cfg so the engine's grid-size gate can
         # fall back to the fused per-member path for tiny populations
         engine = GpuEngine(cfg, comm, policy, nt, env, rs, objective=args.objective,
-                           use_graph=not args.no_graph,
+                           use_graph=not args.no_graph, rollout_mode=args.mode,
                            pair_rollout=False if args.no_pair else None)
         if args.objective == "nsr":
             # seeded starter archive on device (NSR-A semantics: novelty vs

@@ -244,8 +244,7 @@ class GpuEngine:
             want_pair = (bool(cfg.general.get("pair_rollout", False))
                          and self.pairs * self.eps >= 384)
         self.pair_rollout = (want_pair and self.fused and not self.split_dyn
-                             and self.rollout_mode == "step"
-                             and self.steps_per_launch == 1 and self.pairs >= 1)
+                             and self.pairs >= 1)
         self._zero_off = torch.zeros(1, dtype=torch.int64, device=d)
         self._zero_sign = torch.zeros(1, dtype=torch.float32, device=d)
         if self.pair_rollout:
@@ -376,6 +375,37 @@ class GpuEngine:
             float(env.fall_threshold), float(env.dt), self._stream()),
             "es_loco_episode")
 
+    def _loco_pair_episode(self, n_steps: Optional[int] = None, salt_base: int = 0):
+        """n_steps consecutive env steps for every (pair, episode) block in
+        ONE launch (default: the whole episode). Blocks never rendezvous, so
+        dynamics-L2 phases overlap other blocks' HBM weight streaming —
+        the per-step pair grid measured only ~47% HBM duty (phase-locked
+        launches); trajectories are bitwise-identical to per-step launches."""
+        env = self.env
+        goal_ptr = env.goal.data_ptr() if env.goal_conditioned else None
+        ops.check(ops.hip().es_loco_pair_episode(
+            self.theta_row.data_ptr(), self.eps_rows.data_ptr(),
+            self.obmean.data_ptr(), self.obstd.data_ptr(),
+            self.dims_arr.ctypes.data, len(self.dims_arr),
+            self.seed_dev.data_ptr(),
+            float(self.policy._module.ob_clip), self.acstd_dev.data_ptr(),
+            self.row_stride,
+            env.s.data_ptr(), env.pos.data_ptr(), goal_ptr,
+            env.A_bf16.data_ptr(), env.B.data_ptr(), env.b0.data_ptr(),
+            env.wv.data_ptr(), env.wa.data_ptr(), env.wy.data_ptr(),
+            env.wh.data_ptr(),
+            self.alive.data_ptr(), self.rew_total.data_ptr(),
+            self.member_steps.data_ptr(), self.behv.data_ptr(),
+            self.mo_sum.data_ptr(), self.mo_sumsq.data_ptr(),
+            self.pairs, env.sdim, env.ac_dim, int(env.goal_conditioned),
+            int(env.terminate_on_fall), (self.M - 1) * self.eps, self.bins,
+            self.eps, self.act_mode,
+            float(env.leak), float(env.ctrl_cost), float(env.alive_bonus),
+            float(env.fall_threshold), float(env.dt),
+            self.max_steps if n_steps is None else int(n_steps), int(salt_base),
+            int(self.cfg.general.get("pair_block", 256) or 256),
+            self._stream()), "es_loco_pair_episode")
+
     def _loco_noiseless_episode(self):
         if self.pair_rollout:
             self._loco_episode((self.M - 1) * self.eps, self.eps, 0,
@@ -412,7 +442,10 @@ class GpuEngine:
             self._side.wait_stream(main)
             with torch.cuda.stream(self._side):
                 self._loco_noiseless_episode()
-            self._loco_episode(0, (self.M - 1) * self.eps, (self.M - 1) * self.eps)
+            if self.pair_rollout:
+                self._loco_pair_episode()
+            else:
+                self._loco_episode(0, (self.M - 1) * self.eps, (self.M - 1) * self.eps)
             main.wait_stream(self._side)
             return
         if self.fused:
@@ -426,8 +459,11 @@ class GpuEngine:
             if k > 1:
                 mm = (self.M - 1) * self.eps
                 for t0 in range(0, self.max_steps, k):
-                    self._loco_episode(0, mm, mm, n_steps=min(k, self.max_steps - t0),
-                                       salt_base=t0)
+                    n = min(k, self.max_steps - t0)
+                    if self.pair_rollout:
+                        self._loco_pair_episode(n_steps=n, salt_base=t0)
+                    else:
+                        self._loco_episode(0, mm, mm, n_steps=n, salt_base=t0)
             else:
                 for t in range(self.max_steps):
                     self._loco_step(t)

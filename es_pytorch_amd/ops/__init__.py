@@ -183,6 +183,11 @@ def _bind_hip(lib):
                                       p, p, p, p, p, p,
                                       i32, i32, i32, i32, i32, i32, i32, i32, i32,
                                       f32, f32, f32, f32, f32, p]
+    lib.es_loco_pair_episode.argtypes = [p, p, p, p, p, i32, p, f32, p, i64,
+                                         p, p, p, p, p, p, p, p, p, p,
+                                         p, p, p, p, p, p,
+                                         i32, i32, i32, i32, i32, i32, i32, i32, i32,
+                                         f32, f32, f32, f32, f32, i32, i32, i32, p]
     lib.es_loco_episode.argtypes = [p, p, p, p, i32, p, i32, f32, p, i64,
                                     p, p, p, p, p, p, p, p, p, p,
                                     p, p, p, p, p, p,
@@ -190,7 +195,7 @@ def _bind_hip(lib):
                                     i32, i32, f32, f32, f32, f32, f32, p]
     for fn in ["es_noise_fill", "es_pheno_bf16", "es_mlp_fwd", "es_grad_gather",
                "es_adam_step", "es_sgd_step", "es_loco_step", "es_loco_step_split",
-               "es_loco_pair_step", "es_loco_episode"]:
+               "es_loco_pair_step", "es_loco_pair_episode", "es_loco_episode"]:
         getattr(lib, fn).restype = i32
 
 

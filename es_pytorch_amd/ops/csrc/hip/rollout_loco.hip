@@ -482,26 +482,71 @@ __device__ __forceinline__ void loco_pair_step_body(
   loco_dyn_finish(la, P, bm, rawsM, abufM, partial + 2048, tid, nth);
 }
 
+#define ES_LOCO_PAIR_CARVE()                                     \
+  extern __shared__ __attribute__((aligned(16))) char smem[];    \
+  float* bufAp = reinterpret_cast<float*>(smem);                 \
+  float* bufAm = bufAp + sh.maxdim;                              \
+  float* bufBp = bufAm + sh.maxdim;                              \
+  float* bufBm = bufBp + sh.maxdim;                              \
+  float* partial = bufBm + sh.maxdim; /* 2 x 256*8 */            \
+  float* rawsP = partial + 2 * 256 * 8;                          \
+  float* rawsM = rawsP + ((la.S + 3) & ~3);                      \
+  float* abufP = rawsM + ((la.S + 3) & ~3);                      \
+  float* abufM = abufP + 64;                                     \
+  const int q = blockIdx.x;                                      \
+  const int p = q / la.eps, e = q % la.eps;                      \
+  const int bp = p * la.eps + e;                                 \
+  const int bm = (n_pairs + p) * la.eps + e;                     \
+  const uint16_t* ebp = eb + (int64_t)p * la.row_stride;
+
 __global__ void __launch_bounds__(256)
 loco_pair_step_kernel(MlpShape sh, LocoArgs la, LocoPtrs P, const uint16_t* tb,
                       const uint16_t* eb, int n_pairs, uint64_t salt) {
-  extern __shared__ __attribute__((aligned(16))) char smem[];
-  float* bufAp = reinterpret_cast<float*>(smem);
-  float* bufAm = bufAp + sh.maxdim;
-  float* bufBp = bufAm + sh.maxdim;
-  float* bufBm = bufBp + sh.maxdim;
-  float* partial = bufBm + sh.maxdim;  // 2 x 256*8 (forward, then dynamics)
-  float* rawsP = partial + 2 * 256 * 8;
-  float* rawsM = rawsP + ((la.S + 3) & ~3);
-  float* abufP = rawsM + ((la.S + 3) & ~3);
-  float* abufM = abufP + 64;
-  const int q = blockIdx.x;
-  const int p = q / la.eps, e = q % la.eps;
-  const int bp = p * la.eps + e;
-  const int bm = (n_pairs + p) * la.eps + e;
-  loco_pair_step_body(sh, la, P, tb, eb + (int64_t)p * la.row_stride, bp, bm, salt,
-                      bufAp, bufAm, bufBp, bufBm, partial, rawsP, rawsM, abufP,
-                      abufM);
+  ES_LOCO_PAIR_CARVE();
+  loco_pair_step_body(sh, la, P, tb, ebp, bp, bm, salt, bufAp, bufAm, bufBp, bufBm,
+                      partial, rawsP, rawsM, abufP, abufM);
+}
+
+// Whole-episode (or k-step chunk) pair rollout in ONE launch: each block
+// owns its (pair, episode) slot for n_steps consecutive env steps. Unlike
+// the per-step launch, blocks never rendezvous chip-wide, so one block's
+// L2-side dynamics phase overlaps other blocks' HBM sigma*eps streaming
+// (the per-step grid runs phase-locked: every block starts its forward at
+// launch, leaving HBM idle during the correlated dynamics phases — measured
+// 47% HBM duty at the flagship config). Trajectories are bitwise-identical
+// to n_steps sequential es_loco_pair_step launches (same salt_base + t
+// sequence, same body).
+// unroll(disable): the step loop's only carried state is scalar, but an
+// unrolled body doubles the live vector ranges (256 VGPRs = 2 blocks/CU,
+// or 328 B/lane spills under a 3-wave bound). Episode mode REQUIRES
+// >= 3 blocks/CU: 640 flagship blocks must all be resident, since blocks
+// retire only at episode end.
+__global__ void __launch_bounds__(256, 3)
+loco_pair_episode_kernel(MlpShape sh, LocoArgs la, LocoPtrs P, const uint16_t* tb,
+                         const uint16_t* eb, int n_pairs, int n_steps,
+                         int salt_base) {
+  ES_LOCO_PAIR_CARVE();
+#pragma clang loop unroll(disable)
+  for (int t = 1; t <= n_steps; ++t)
+    loco_pair_step_body(sh, la, P, tb, ebp, bp, bm, (uint64_t)(salt_base + t),
+                        bufAp, bufAm, bufBp, bufBm, partial, rawsP, rawsM, abufP,
+                        abufM);
+}
+
+// 128-thread variant: 2 waves/SIMD at the full 256-VGPR budget -> zero
+// spills and 4 resident blocks/CU (8 waves). Trades wave count for clean
+// registers; which wins is measured, not assumed (tools/kbench.py --pair
+// --episode --bs 128 vs 256).
+__global__ void __launch_bounds__(128, 2)
+loco_pair_episode128_kernel(MlpShape sh, LocoArgs la, LocoPtrs P, const uint16_t* tb,
+                            const uint16_t* eb, int n_pairs, int n_steps,
+                            int salt_base) {
+  ES_LOCO_PAIR_CARVE();
+#pragma clang loop unroll(disable)
+  for (int t = 1; t <= n_steps; ++t)
+    loco_pair_step_body(sh, la, P, tb, ebp, bp, bm, (uint64_t)(salt_base + t),
+                        bufAp, bufAm, bufBp, bufBm, partial, rawsP, rawsM, abufP,
+                        abufM);
 }
 
 static int loco_prepare(MlpShape* sh, LocoArgs* la, const int32_t* dims_host, int32_t ndims,
@@ -646,6 +691,47 @@ extern "C" int es_loco_pair_step(
   const unsigned grid = (unsigned)(n_pairs * la.eps);
   loco_pair_step_kernel<<<dim3(grid), dim3(256), lds, (hipStream_t)stream>>>(
       sh, la, P, (const uint16_t*)theta_row, (const uint16_t*)eps_rows, n_pairs, salt);
+  ES_CHECK_LAUNCH();
+  return 0;
+}
+
+// Pair-episode: n_steps consecutive env steps per launch (n_steps =
+// max_steps -> whole generation in one launch; smaller -> chunked step
+// mode). Same buffers as es_loco_pair_step.
+extern "C" int es_loco_pair_episode(
+    const void* theta_row, const void* eps_rows, const void* obmean,
+    const void* obstd, const int32_t* dims_host, int32_t ndims, const void* seed_dev,
+    float ob_clip, const void* ac_std_dev, int64_t row_stride,
+    void* s_glob, void* pos, const void* goal, const void* Am, const void* Bm,
+    const void* b0, const void* wv, const void* wa, const void* wy, const void* wh,
+    void* alive, void* rew_total, void* member_steps, void* behv, void* mo_sum,
+    void* mo_sumsq, int32_t n_pairs, int32_t sdim, int32_t adim, int32_t goal_flag,
+    int32_t terminate, int32_t noiseless_from, int32_t bins, int32_t eps,
+    int32_t act_mode, float leak, float ctrl, float alive_bonus, float fall_thr,
+    float dt, int32_t n_steps, int32_t salt_base, int32_t block_threads,
+    void* stream) {
+  MlpShape sh;
+  LocoArgs la;
+  unsigned lds_unused;
+  int rc = loco_prepare(&sh, &la, dims_host, ndims, row_stride, ob_clip, sdim, adim,
+                        goal_flag, terminate, noiseless_from, bins, eps, act_mode, leak,
+                        ctrl, alive_bonus, fall_thr, dt, &lds_unused);
+  if (rc) return rc;
+  LocoPtrs P = loco_ptrs(nullptr, obmean, obstd, ac_std_dev, seed_dev, s_glob, pos,
+                         goal, Am, Bm, b0, wv, wa, wy, wh, alive, rew_total,
+                         member_steps, behv, mo_sum, mo_sumsq);
+  const int Spad = (sdim + 3) & ~3;
+  const unsigned lds =
+      (unsigned)((4 * sh.maxdim + 2 * 256 * 8 + 2 * Spad + 2 * 64) * sizeof(float));
+  const unsigned grid = (unsigned)(n_pairs * la.eps);
+  if (block_threads == 128)
+    loco_pair_episode128_kernel<<<dim3(grid), dim3(128), lds, (hipStream_t)stream>>>(
+        sh, la, P, (const uint16_t*)theta_row, (const uint16_t*)eps_rows, n_pairs,
+        n_steps, salt_base);
+  else
+    loco_pair_episode_kernel<<<dim3(grid), dim3(256), lds, (hipStream_t)stream>>>(
+        sh, la, P, (const uint16_t*)theta_row, (const uint16_t*)eps_rows, n_pairs,
+        n_steps, salt_base);
   ES_CHECK_LAUNCH();
   return 0;
 }
