@@ -53,8 +53,6 @@ def main():
                    help="rollout launch shape: per-step graph-replayed grid, or "
                         "one whole-episode launch per generation (blocks drift; "
                         "phases overlap chip-wide)")
-    p.add_argument("--pair-block", type=int, default=256, choices=[128, 256],
-                   help="pair-episode kernel block size")
     args = p.parse_args()
 
     from es_pytorch_amd.config import AttrDict
@@ -89,8 +87,7 @@ def main():
         "policy": {"layer_sizes": list(args.layers), "ac_std": 0.01, "l2coeff": 0.005,
                    "lr": 0.01, "ob_clip": 5, "save_obs_chance": 0.01},
         "general": {"name": "bench", "policies_per_gen": ppg, "batch_size": 500,
-                    "seed": global_seed, "pair_rollout": not args.no_pair,
-                    "pair_block": args.pair_block},
+                    "seed": global_seed, "pair_rollout": not args.no_pair},
     })
 
     B = 2 * (ppg // world // 2) + 1

@@ -248,10 +248,15 @@ class GpuEngine:
         self._zero_off = torch.zeros(1, dtype=torch.int64, device=d)
         self._zero_sign = torch.zeros(1, dtype=torch.float32, device=d)
         if self.pair_rollout:
-            self.theta_row = torch.empty((1, self.row_stride), dtype=torch.bfloat16,
-                                         device=d)
-            self.eps_rows = torch.empty((self.pairs, self.row_stride),
-                                        dtype=torch.bfloat16, device=d)
+            # +1 guard row: the pair forward's prefetch ring may overrun the
+            # final layer's weights by up to DE*PART rows (values are never
+            # consumed, but the addresses must stay inside the allocation)
+            self._theta_blob = torch.empty((2, self.row_stride), dtype=torch.bfloat16,
+                                           device=d)
+            self.theta_row = self._theta_blob[:1]
+            self._eps_blob = torch.empty((self.pairs + 1, self.row_stride),
+                                         dtype=torch.bfloat16, device=d)
+            self.eps_rows = self._eps_blob[:self.pairs]
             self._zeros_n = torch.zeros(self.n, dtype=torch.float32, device=d)
             self._one_signs = torch.ones(self.pairs, dtype=torch.float32, device=d)
         self._warned_host_ranker = False
@@ -403,7 +408,6 @@ class GpuEngine:
             float(env.leak), float(env.ctrl_cost), float(env.alive_bonus),
             float(env.fall_threshold), float(env.dt),
             self.max_steps if n_steps is None else int(n_steps), int(salt_base),
-            int(self.cfg.general.get("pair_block", 256) or 256),
             self._stream()), "es_loco_pair_episode")
 
     def _loco_noiseless_episode(self):

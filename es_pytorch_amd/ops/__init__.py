@@ -187,7 +187,7 @@ def _bind_hip(lib):
                                          p, p, p, p, p, p, p, p, p, p,
                                          p, p, p, p, p, p,
                                          i32, i32, i32, i32, i32, i32, i32, i32, i32,
-                                         f32, f32, f32, f32, f32, i32, i32, i32, p]
+                                         f32, f32, f32, f32, f32, i32, i32, p]
     lib.es_loco_episode.argtypes = [p, p, p, p, i32, p, i32, f32, p, i64,
                                     p, p, p, p, p, p, p, p, p, p,
                                     p, p, p, p, p, p,

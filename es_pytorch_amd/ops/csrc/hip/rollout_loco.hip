@@ -521,7 +521,7 @@ loco_pair_step_kernel(MlpShape sh, LocoArgs la, LocoPtrs P, const uint16_t* tb,
 // or 328 B/lane spills under a 3-wave bound). Episode mode REQUIRES
 // >= 3 blocks/CU: 640 flagship blocks must all be resident, since blocks
 // retire only at episode end.
-__global__ void __launch_bounds__(256, 3)
+__global__ void __launch_bounds__(256, 4)
 loco_pair_episode_kernel(MlpShape sh, LocoArgs la, LocoPtrs P, const uint16_t* tb,
                          const uint16_t* eb, int n_pairs, int n_steps,
                          int salt_base) {
@@ -533,21 +533,9 @@ loco_pair_episode_kernel(MlpShape sh, LocoArgs la, LocoPtrs P, const uint16_t* t
                         abufM);
 }
 
-// 128-thread variant: 2 waves/SIMD at the full 256-VGPR budget -> zero
-// spills and 4 resident blocks/CU (8 waves). Trades wave count for clean
-// registers; which wins is measured, not assumed (tools/kbench.py --pair
-// --episode --bs 128 vs 256).
-__global__ void __launch_bounds__(128, 2)
-loco_pair_episode128_kernel(MlpShape sh, LocoArgs la, LocoPtrs P, const uint16_t* tb,
-                            const uint16_t* eb, int n_pairs, int n_steps,
-                            int salt_base) {
-  ES_LOCO_PAIR_CARVE();
-#pragma clang loop unroll(disable)
-  for (int t = 1; t <= n_steps; ++t)
-    loco_pair_step_body(sh, la, P, tb, ebp, bp, bm, (uint64_t)(salt_base + t),
-                        bufAp, bufAm, bufBp, bufBm, partial, rawsP, rawsM, abufP,
-                        abufM);
-}
+// (A 128-thread 2-waves/SIMD variant — zero spills, 4 blocks/CU — was
+// measured 24% SLOWER than per-step launches and cannot match them bitwise
+// (different PART split changes summation order); removed after the A/B.)
 
 static int loco_prepare(MlpShape* sh, LocoArgs* la, const int32_t* dims_host, int32_t ndims,
                         int64_t row_stride, float ob_clip, int32_t sdim, int32_t adim,
@@ -708,8 +696,7 @@ extern "C" int es_loco_pair_episode(
     void* mo_sumsq, int32_t n_pairs, int32_t sdim, int32_t adim, int32_t goal_flag,
     int32_t terminate, int32_t noiseless_from, int32_t bins, int32_t eps,
     int32_t act_mode, float leak, float ctrl, float alive_bonus, float fall_thr,
-    float dt, int32_t n_steps, int32_t salt_base, int32_t block_threads,
-    void* stream) {
+    float dt, int32_t n_steps, int32_t salt_base, void* stream) {
   MlpShape sh;
   LocoArgs la;
   unsigned lds_unused;
@@ -724,14 +711,9 @@ extern "C" int es_loco_pair_episode(
   const unsigned lds =
       (unsigned)((4 * sh.maxdim + 2 * 256 * 8 + 2 * Spad + 2 * 64) * sizeof(float));
   const unsigned grid = (unsigned)(n_pairs * la.eps);
-  if (block_threads == 128)
-    loco_pair_episode128_kernel<<<dim3(grid), dim3(128), lds, (hipStream_t)stream>>>(
-        sh, la, P, (const uint16_t*)theta_row, (const uint16_t*)eps_rows, n_pairs,
-        n_steps, salt_base);
-  else
-    loco_pair_episode_kernel<<<dim3(grid), dim3(256), lds, (hipStream_t)stream>>>(
-        sh, la, P, (const uint16_t*)theta_row, (const uint16_t*)eps_rows, n_pairs,
-        n_steps, salt_base);
+  loco_pair_episode_kernel<<<dim3(grid), dim3(256), lds, (hipStream_t)stream>>>(
+      sh, la, P, (const uint16_t*)theta_row, (const uint16_t*)eps_rows, n_pairs,
+      n_steps, salt_base);
   ES_CHECK_LAUNCH();
   return 0;
 }

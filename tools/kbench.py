@@ -31,8 +31,6 @@ def main():
     p.add_argument("--pair", action="store_true", help="antithetic-pair rollout")
     p.add_argument("--dyn-group", type=int, default=5,
                    help="members per dynamics block in split mode")
-    p.add_argument("--bs", type=int, default=256, choices=[128, 256],
-                   help="pair-episode kernel block size")
     args = p.parse_args()
 
     from es_pytorch_amd.config import AttrDict
@@ -54,8 +52,7 @@ def main():
         "policy": {"layer_sizes": list(args.layers), "ac_std": 0.01, "l2coeff": 0.005,
                    "lr": 0.01, "ob_clip": 5, "save_obs_chance": 0.01},
         "general": {"policies_per_gen": args.pop, "batch_size": 500, "seed": 5,
-                    "steps_per_launch": args.chunk, "dyn_group": args.dyn_group,
-                    "pair_block": args.bs},
+                    "steps_per_launch": args.chunk, "dyn_group": args.dyn_group},
     })
     B = args.pop + 1
     env = make_batched(args.env, B, dev, max_steps=args.steps, terminate_on_fall=False)

@@ -12,6 +12,7 @@ the exchange payload is rank-dependent, the merged result must not be).
 
 Exit 0 = all invariants held on every rank.
 """
+import hashlib
 import os
 import sys
 
@@ -58,7 +59,7 @@ def main():
     ranker = CenteredRanker()
 
     # rank-distinct offset streams, rank-identical initial params
-    h0 = comm.allgather_obj(hash(policy.flat_params.tobytes()))
+    h0 = comm.allgather_obj(hashlib.sha256(policy.flat_params.tobytes()).hexdigest())
     assert len(set(h0)) == 1, f"initial params differ across ranks: {h0}"
 
     for gen in range(4):
@@ -67,7 +68,7 @@ def main():
         assert ranker.n_fits_ranked == ppg, (ranker.n_fits_ranked, ppg)
         # bitwise-identical parameters on every rank after the redundant update
         theta_bytes = eng.theta.cpu().numpy().tobytes()
-        hashes = comm.allgather_obj(hash(theta_bytes))
+        hashes = comm.allgather_obj(hashlib.sha256(theta_bytes).hexdigest())
         assert len(set(hashes)) == 1, f"gen {gen}: rank params diverged: {hashes}"
         # distinct noise draws per rank (the exchange is not degenerate)
         offs = comm.allgather_obj(int(eng.offsets[0].item()))
