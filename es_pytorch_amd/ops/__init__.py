@@ -23,7 +23,9 @@ from typing import Optional
 
 _DIR = os.path.dirname(os.path.abspath(__file__))
 _CPU_SO = os.path.join(_DIR, "_cpu_ops.so")
-_HIP_SO = os.path.join(_DIR, "_hip_ops.so")
+# ES_HIP_SO: load an alternative prebuilt kernel library (A/B experiments
+# with different -DES_DEPTH_* builds on one box; see tools/build_variants.py)
+_HIP_SO = os.environ.get("ES_HIP_SO", os.path.join(_DIR, "_hip_ops.so"))
 
 _cpu_lib: Optional[ctypes.CDLL] = None
 _hip_lib: Optional[ctypes.CDLL] = None
@@ -88,28 +90,30 @@ def hip_sources():
     return sorted(os.path.join(d, f) for f in os.listdir(d) if f.endswith(".hip"))
 
 
-def build_hip(force: bool = False, arch: str = "gfx950") -> str:
+def build_hip(force: bool = False, arch: str = "gfx950", extra_flags=None,
+              out: Optional[str] = None) -> str:
     srcs = hip_sources()
     hdrs = [os.path.join(_DIR, "csrc", "philox.h")] + \
         [os.path.join(_DIR, "csrc", "hip", h)
          for h in os.listdir(os.path.join(_DIR, "csrc", "hip")) if h.endswith(".h")]
     newest = max(os.path.getmtime(f) for f in srcs + hdrs)
+    target = out or _HIP_SO
 
     def stale():
-        return force or not os.path.exists(_HIP_SO) or os.path.getmtime(_HIP_SO) < newest
+        return force or not os.path.exists(target) or os.path.getmtime(target) < newest
 
     if stale():
-        with _BuildLock(_HIP_SO):
+        with _BuildLock(target):
             if stale():  # re-check under the lock: another rank may have built
                 hipcc = os.environ.get("HIPCC", "/opt/rocm/bin/hipcc")
-                tmp = _HIP_SO + ".tmp"
+                tmp = target + ".tmp"
                 cmd = [hipcc, f"--offload-arch={arch}", "-O3", "-std=c++17", "-shared",
-                       "-fPIC", *srcs, "-o", tmp]
+                       "-fPIC", *(extra_flags or []), *srcs, "-o", tmp]
                 r = subprocess.run(cmd, check=False, capture_output=True, text=True)
                 if r.returncode != 0:
                     raise RuntimeError(f"hipcc build failed:\n{r.stdout}\n{r.stderr}")
-                os.replace(tmp, _HIP_SO)
-    return _HIP_SO
+                os.replace(tmp, target)
+    return target
 
 
 def cpu() -> ctypes.CDLL:

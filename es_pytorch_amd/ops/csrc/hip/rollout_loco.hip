@@ -499,7 +499,12 @@ __device__ __forceinline__ void loco_pair_step_body(
   const int bm = (n_pairs + p) * la.eps + e;                     \
   const uint16_t* ebp = eb + (int64_t)p * la.row_stride;
 
-__global__ void __launch_bounds__(256)
+// ES_PAIR_MINWAVES (experiment knob): force a min-waves/SIMD bound on the
+// pair step kernel so deeper rings can be capped to 3 waves (168 VGPRs)
+#ifndef ES_PAIR_MINWAVES
+#define ES_PAIR_MINWAVES 1
+#endif
+__global__ void __launch_bounds__(256, ES_PAIR_MINWAVES)
 loco_pair_step_kernel(MlpShape sh, LocoArgs la, LocoPtrs P, const uint16_t* tb,
                       const uint16_t* eb, int n_pairs, uint64_t salt) {
   ES_LOCO_PAIR_CARVE();
