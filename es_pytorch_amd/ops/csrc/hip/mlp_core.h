@@ -13,18 +13,6 @@
 #define ES_MAXL 8
 #define ES_MAXDIM 2048
 
-// Pair-forward register-ring depths (overridable with -D for experiments):
-// eps = HBM stream (~900 cyc misses), theta = L2-resident (~200 cyc).
-// Allocation is non-monotonic in ring size (measured, gfx950 hipcc):
-// (4,4) -> 128 VGPRs = 4 waves/SIMD = 4 blocks/CU (the sweet spot);
-// (2,2) 147, (4,2) 148, (8,4) 170, (8,8) 195, (8,2) 238.
-#ifndef ES_DEPTH_E
-#define ES_DEPTH_E 4
-#endif
-#ifndef ES_DEPTH_T
-#define ES_DEPTH_T 4
-#endif
-
 struct MlpShape {
   int n_layers;
   int dims[ES_MAXL + 1];
@@ -282,44 +270,30 @@ __device__ __forceinline__ void mlp_layers_pair(
         auto ldt = [&](int i) {
           return *reinterpret_cast<const uint4*>(tcol + (int64_t)i * O);
         };
-        // Asymmetric-depth register rings: the sigma*eps stream is an HBM
-        // miss every element (~900 cyc) and gets ES_DEPTH_E loads of lead;
-        // theta is L2/MALL-resident (~200 cyc) and needs only ES_DEPTH_T.
-        // Deeper eps lead at FEWER live registers than the old symmetric
-        // 8+8 double-buffer. Ring indices stay compile-time because the
-        // chunk width CH is a multiple of both depths. Loads may overrun
-        // the layer (never the +1 guard row the engine allocates); overrun
-        // values are never consumed. Accumulation order (ascending i) is
-        // unchanged -> bitwise-identical results.
-        constexpr int DE = ES_DEPTH_E, DT = ES_DEPTH_T, CH = 8;
-        static_assert(CH % DE == 0 || DE % CH == 0, "ring phase");
-        static_assert(CH % DT == 0, "ring phase");
         int i = ip;
-        if (DE <= CH && i + (CH - 1) * PART < I) {
-          uint4 er[DE], tr[DT];
-#pragma unroll
-          for (int j = 0; j < DE; ++j) er[j] = lde(i + j * PART);
-#pragma unroll
-          for (int j = 0; j < DT; ++j) tr[j] = ldt(i + j * PART);
-          for (; i + (CH - 1) * PART < I; i += CH * PART) {
-#pragma unroll
-            for (int j = 0; j < CH; ++j) {
-              const uint4 ce = er[j % DE], ct = tr[j % DT];
-              er[j % DE] = lde(i + (j + DE) * PART);
-              tr[j % DT] = ldt(i + (j + DT) * PART);
-              bf8_fma_pair(ct, ce, xp[i + j * PART], xm[i + j * PART], accp, accm);
-            }
+        const int step4 = PART * 4;
+        if (i + 3 * PART < I) {
+          uint4 t0 = ldt(i), t1 = ldt(i + PART), t2 = ldt(i + 2 * PART),
+                t3 = ldt(i + 3 * PART);
+          uint4 e0 = lde(i), e1 = lde(i + PART), e2 = lde(i + 2 * PART),
+                e3 = lde(i + 3 * PART);
+          for (; i + 7 * PART < I; i += step4) {
+            const uint4 nt0 = ldt(i + 4 * PART), nt1 = ldt(i + 5 * PART),
+                        nt2 = ldt(i + 6 * PART), nt3 = ldt(i + 7 * PART);
+            const uint4 ne0 = lde(i + 4 * PART), ne1 = lde(i + 5 * PART),
+                        ne2 = lde(i + 6 * PART), ne3 = lde(i + 7 * PART);
+            bf8_fma_pair(t0, e0, xp[i], xm[i], accp, accm);
+            bf8_fma_pair(t1, e1, xp[i + PART], xm[i + PART], accp, accm);
+            bf8_fma_pair(t2, e2, xp[i + 2 * PART], xm[i + 2 * PART], accp, accm);
+            bf8_fma_pair(t3, e3, xp[i + 3 * PART], xm[i + 3 * PART], accp, accm);
+            t0 = nt0; t1 = nt1; t2 = nt2; t3 = nt3;
+            e0 = ne0; e1 = ne1; e2 = ne2; e3 = ne3;
           }
-          // drain the ring's already-loaded leading blocks (no new loads)
-          int consumed = 0;
-#pragma unroll
-          for (int j = 0; j < DT; ++j)
-            if (i + j * PART < I) {
-              bf8_fma_pair(tr[j % DT], er[j % DE], xp[i + j * PART],
-                           xm[i + j * PART], accp, accm);
-              consumed = j + 1;
-            }
-          i += consumed * PART;
+          bf8_fma_pair(t0, e0, xp[i], xm[i], accp, accm);
+          bf8_fma_pair(t1, e1, xp[i + PART], xm[i + PART], accp, accm);
+          bf8_fma_pair(t2, e2, xp[i + 2 * PART], xm[i + 2 * PART], accp, accm);
+          bf8_fma_pair(t3, e3, xp[i + 3 * PART], xm[i + 3 * PART], accp, accm);
+          i += step4;
         }
         for (; i < I; i += PART)
           bf8_fma_pair(ldt(i), lde(i), xp[i], xm[i], accp, accm);
