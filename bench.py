@@ -49,6 +49,10 @@ def main():
                         "measured 1.27x over per-member weight blobs, and "
                         "the perturbation keeps MORE bf16 mantissa)")
     p.add_argument("--cpu", action="store_true", help="debug: run the engine on CPU")
+    p.add_argument("--eps-fp8", action="store_true",
+                   help="fp8 (e4m3) sigma*eps stream: halves the dominant "
+                        "HBM weight bytes; fitness fidelity documented in "
+                        "profiles/pair_fidelity_sweep.txt")
     p.add_argument("--mode", type=str, default="step", choices=["step", "episode"],
                    help="rollout launch shape: per-step graph-replayed grid, or "
                         "one whole-episode launch per generation (blocks drift; "
@@ -112,7 +116,8 @@ def main():
         # fall back to the fused per-member path for tiny populations
         engine = GpuEngine(cfg, comm, policy, nt, env, rs, objective=args.objective,
                            use_graph=not args.no_graph, rollout_mode=args.mode,
-                           pair_rollout=False if args.no_pair else None)
+                           pair_rollout=False if args.no_pair else None,
+                           eps_fp8=args.eps_fp8 or None)
         if args.objective == "nsr":
             # seeded starter archive on device (NSR-A semantics: novelty vs
             # the behaviour archive, grown per generation)

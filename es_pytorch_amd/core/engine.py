@@ -82,7 +82,8 @@ class GpuEngine:
                  use_graph: bool = True, novelty_k: int = 10,
                  fused: Optional[bool] = None, rollout_mode: str = "step",
                  split_dyn: Optional[bool] = None,
-                 pair_rollout: Optional[bool] = None):
+                 pair_rollout: Optional[bool] = None,
+                 eps_fp8: Optional[bool] = None):
         # rollout_mode: "step" = one kernel per env step for the population
         # (graph-replayed); "episode" = ONE kernel per generation, each block
         # runs its member's whole episode (members are mutually independent,
@@ -247,19 +248,47 @@ class GpuEngine:
                              and self.pairs >= 1)
         self._zero_off = torch.zeros(1, dtype=torch.int64, device=d)
         self._zero_sign = torch.zeros(1, dtype=torch.float32, device=d)
+        # fp8 sigma*eps stream (halves the dominant HBM bytes AND load count;
+        # OCP e4m3fn, exact antithetic cancellation): gated on the pair path,
+        # an explicit opt-in, and the row-pair-interleave layout constraints
+        # (even input dim + 16-B-aligned weight offsets on vec layers)
+        want_fp8 = (bool(cfg.general.get("eps_fp8", False))
+                    if eps_fp8 is None else bool(eps_fp8))
+        self.eps_fp8 = (want_fp8 and self.pair_rollout
+                        and self.rollout_mode == "step"
+                        and self.steps_per_launch == 1
+                        and self._fp8_layout_ok(self.dims))
+        if want_fp8 and not self.eps_fp8 and comm.rank == 0:
+            import sys
+            print("[engine] eps_fp8 requested but unavailable for this "
+                  "model/path — using bf16 eps rows", file=sys.stderr)
         if self.pair_rollout:
-            # +1 guard row: the pair forward's prefetch ring may overrun the
-            # final layer's weights by up to DE*PART rows (values are never
-            # consumed, but the addresses must stay inside the allocation)
+            # +1 guard row: the pair forward's prefetch pipeline may overrun
+            # the final layer's weights (values are never consumed, but the
+            # addresses must stay inside the allocation)
             self._theta_blob = torch.empty((2, self.row_stride), dtype=torch.bfloat16,
                                            device=d)
             self.theta_row = self._theta_blob[:1]
+            eps_dt = torch.uint8 if self.eps_fp8 else torch.bfloat16
             self._eps_blob = torch.empty((self.pairs + 1, self.row_stride),
-                                         dtype=torch.bfloat16, device=d)
+                                         dtype=eps_dt, device=d)
             self.eps_rows = self._eps_blob[:self.pairs]
             self._zeros_n = torch.zeros(self.n, dtype=torch.float32, device=d)
             self._one_signs = torch.ones(self.pairs, dtype=torch.float32, device=d)
         self._warned_host_ranker = False
+
+    @staticmethod
+    def _fp8_layout_ok(dims: List[int]) -> bool:
+        """Mirror of the fp8 row-pair interleave constraints (pheno.hip):
+        every vectorizable layer needs an even input dim and a 16-byte-
+        aligned weight offset so the 16-fp8 loads stay aligned."""
+        off = 0
+        for I, O in zip(dims[:-1], dims[1:]):
+            vec = (O % 8 == 0) and (off % 8 == 0)
+            if vec and (I % 2 != 0 or off % 16 != 0):
+                return False
+            off += I * O + O
+        return True
 
     # ------------------------------------------------------------------ ops
     def _stream(self):
@@ -275,6 +304,14 @@ class GpuEngine:
                 self.nt.noise.data_ptr(), self._zero_off.data_ptr(),
                 self._zero_sign.data_ptr(), 1, self.n, self.row_stride, 0.0,
                 self._stream()), "es_pheno_bf16")
+            if self.eps_fp8:
+                # sigma*eps rows in e4m3, row-pair-interleaved (pheno.hip)
+                ops.check(ops.hip().es_pheno_fp8(
+                    self.eps_rows.data_ptr(), self.nt.noise.data_ptr(),
+                    self.offsets.data_ptr(), self.dims_arr.ctypes.data,
+                    len(self.dims_arr), self.pairs, self.n, self.row_stride,
+                    std, self._stream()), "es_pheno_fp8")
+                return
             # sigma*eps rows: zero theta + sign +1 -> bf16(sigma*noise[off_p:])
             ops.check(ops.hip().es_pheno_bf16(
                 self.eps_rows.data_ptr(), self._zeros_n.data_ptr(),
@@ -307,7 +344,9 @@ class GpuEngine:
         env = self.env
         goal_ptr = env.goal.data_ptr() if env.goal_conditioned else None
         if self.pair_rollout:
-            ops.check(ops.hip().es_loco_pair_step(
+            fn = ops.hip().es_loco_pair_step_fp8 if self.eps_fp8 else \
+                ops.hip().es_loco_pair_step
+            ops.check(fn(
                 self.theta_row.data_ptr(), self.eps_rows.data_ptr(),
                 self.obmean.data_ptr(), self.obstd.data_ptr(),
                 self.dims_arr.ctypes.data, len(self.dims_arr),
@@ -326,7 +365,7 @@ class GpuEngine:
                 self.eps, self.act_mode,
                 float(env.leak), float(env.ctrl_cost), float(env.alive_bonus),
                 float(env.fall_threshold), float(env.dt), self._stream()),
-                "es_loco_pair_step")
+                "es_loco_pair_step_fp8" if self.eps_fp8 else "es_loco_pair_step")
             return
         common = (
             self.weights.data_ptr(), self.obmean.data_ptr(), self.obstd.data_ptr(),

@@ -167,6 +167,7 @@ def _bind_hip(lib):
                                   ctypes.c_int32, ctypes.c_float, ctypes.c_void_p)
     lib.es_noise_fill.argtypes = [p, i64, u64, u32, p]
     lib.es_pheno_bf16.argtypes = [p, p, p, p, p, i64, i64, i64, f32, p]
+    lib.es_pheno_fp8.argtypes = [p, p, p, p, i32, i64, i64, i64, f32, p]
     lib.es_mlp_fwd.argtypes = [p, p, p, p, p, p, i32, p, u64, i32, f32, p, i64, i32,
                                i32, i32, i32, i32, p, p, p]
     lib.es_grad_gather.argtypes = [p, p, p, p, i64, i64, p]
@@ -187,6 +188,7 @@ def _bind_hip(lib):
                                       p, p, p, p, p, p,
                                       i32, i32, i32, i32, i32, i32, i32, i32, i32,
                                       f32, f32, f32, f32, f32, p]
+    lib.es_loco_pair_step_fp8.argtypes = lib.es_loco_pair_step.argtypes
     lib.es_loco_pair_episode.argtypes = [p, p, p, p, p, i32, p, f32, p, i64,
                                          p, p, p, p, p, p, p, p, p, p,
                                          p, p, p, p, p, p,
@@ -197,9 +199,10 @@ def _bind_hip(lib):
                                     p, p, p, p, p, p,
                                     i32, i32, i32, i32, i32, i32, i32, i32, i32, i32,
                                     i32, i32, f32, f32, f32, f32, f32, p]
-    for fn in ["es_noise_fill", "es_pheno_bf16", "es_mlp_fwd", "es_grad_gather",
-               "es_adam_step", "es_sgd_step", "es_loco_step", "es_loco_step_split",
-               "es_loco_pair_step", "es_loco_pair_episode", "es_loco_episode"]:
+    for fn in ["es_noise_fill", "es_pheno_bf16", "es_pheno_fp8", "es_mlp_fwd",
+               "es_grad_gather", "es_adam_step", "es_sgd_step", "es_loco_step",
+               "es_loco_step_split", "es_loco_pair_step", "es_loco_pair_step_fp8",
+               "es_loco_pair_episode", "es_loco_episode"]:
         getattr(lib, fn).restype = i32
 
 

@@ -32,3 +32,29 @@ __device__ __forceinline__ uint16_t f2bf(float f) {
 __device__ __forceinline__ float fclampf(float v, float lo, float hi) {
   return fminf(fmaxf(v, lo), hi);
 }
+
+// fp8 (OCP e4m3fn on gfx950) pack/unpack via the native cvt instructions —
+// encode and decode use the same hardware, so the round trip is
+// self-consistent by construction.
+typedef float es_f32x2 __attribute__((ext_vector_type(2)));
+
+__device__ __forceinline__ void fp8x4_decode(uint32_t w, float* out4) {
+  es_f32x2 lo = __builtin_amdgcn_cvt_pk_f32_fp8(w, false);
+  es_f32x2 hi = __builtin_amdgcn_cvt_pk_f32_fp8(w, true);
+  out4[0] = lo.x;
+  out4[1] = lo.y;
+  out4[2] = hi.x;
+  out4[3] = hi.y;
+}
+
+__device__ __forceinline__ uint32_t fp8x4_encode(float a, float b, float c, float d) {
+  uint32_t w = 0;
+  w = __builtin_amdgcn_cvt_pk_fp8_f32(a, b, w, false);
+  w = __builtin_amdgcn_cvt_pk_fp8_f32(c, d, w, true);
+  return w;
+}
+
+__device__ __forceinline__ float fp8_byte(uint8_t b) {
+  es_f32x2 lo = __builtin_amdgcn_cvt_pk_f32_fp8((uint32_t)b, false);
+  return lo.x;
+}

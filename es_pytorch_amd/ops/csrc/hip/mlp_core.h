@@ -191,6 +191,171 @@ __device__ __forceinline__ float* mlp_layers(const uint16_t* __restrict__ wb,
   return x;
 }
 
+// theta octet (bf16) +- decoded fp8 eps octet, FMA'd into both members
+__device__ __forceinline__ void bf8e_fma_pair(uint4 t, const float* e8, float xp,
+                                              float xm, float* accp, float* accm) {
+  const uint32_t ts[4] = {t.x, t.y, t.z, t.w};
+#pragma unroll
+  for (int q = 0; q < 4; ++q) {
+    const float t0 = bf2f((uint16_t)(ts[q] & 0xFFFFu));
+    const float t1 = bf2f((uint16_t)(ts[q] >> 16));
+    accp[2 * q] = fmaf(t0 + e8[2 * q], xp, accp[2 * q]);
+    accm[2 * q] = fmaf(t0 - e8[2 * q], xm, accm[2 * q]);
+    accp[2 * q + 1] = fmaf(t1 + e8[2 * q + 1], xp, accp[2 * q + 1]);
+    accm[2 * q + 1] = fmaf(t1 - e8[2 * q + 1], xm, accm[2 * q + 1]);
+  }
+}
+
+// mlp_layers_pair with an fp8 (OCP e4m3fn) sigma*eps blob: one nt 16-B load
+// covers one octet of TWO consecutive i-rows (pheno_fp8_kernel's row-pair
+// interleave), halving both the bytes and the load count of the HBM eps
+// stream; theta stays bf16 (L2-resident). Same (oi, ip) tiling; the i-walk
+// is by row PAIRS, so the per-thread accumulation partition differs from
+// the bf16 path (documented; the fp8 path is new numerics regardless).
+__device__ __forceinline__ void mlp_layers_pair_fp8(
+    const uint16_t* __restrict__ tb, const uint8_t* __restrict__ eb8,
+    const MlpShape& sh, float* bufAp, float* bufAm, float* bufBp, float* bufBm,
+    float* partial, int tid, int nthreads, int act_final,
+    float** xp_out, float** xm_out) {
+  float* xp = bufAp;
+  float* xm = bufAm;
+  float* yp = bufBp;
+  float* ym = bufBm;
+  float* partm = partial + 256 * 8;
+  for (int l = 0; l < sh.n_layers; ++l) {
+    const int I = sh.dims[l], O = sh.dims[l + 1];
+    const uint16_t* Tt = tb + sh.woff[l];
+    const uint8_t* Et8 = eb8 + sh.woff[l];
+    const uint16_t* TBs = tb + sh.boff[l];
+    const uint8_t* EBs8 = eb8 + sh.boff[l];
+    const bool do_act = (l < sh.n_layers - 1) || act_final;
+
+    if (sh.vec_ok[l] && !(I & 1)) {
+      const int OCT = O >> 3;
+      const int PART = nthreads / OCT;
+      const int oi = tid % OCT, ip = tid / OCT;
+      float accp[8], accm[8];
+#pragma unroll
+      for (int q = 0; q < 8; ++q) accp[q] = accm[q] = 0.0f;
+      if (ip < PART) {
+        const int I2 = I >> 1;           // row pairs
+        const int rs2 = 2 * O;           // bytes per row pair
+        const uint16_t* tcol = Tt + (oi << 3);
+        const uint8_t* ecol8 = Et8 + (oi << 4);
+        typedef uint32_t u32x4v __attribute__((ext_vector_type(4)));
+        auto lde8 = [&](int i2) {  // 16 fp8 = octet of rows 2*i2, 2*i2+1
+          u32x4v v = __builtin_nontemporal_load(
+              reinterpret_cast<const u32x4v*>(ecol8 + (int64_t)i2 * rs2));
+          uint4 w;
+          w.x = v.x; w.y = v.y; w.z = v.z; w.w = v.w;
+          return w;
+        };
+        auto ldt = [&](int r) {
+          return *reinterpret_cast<const uint4*>(tcol + (int64_t)r * O);
+        };
+        auto fma_pairblk = [&](const uint4& e16, const uint4& ta, const uint4& tb_,
+                               int r0) {
+          float e8[8];
+          fp8x4_decode(e16.x, e8);
+          fp8x4_decode(e16.y, e8 + 4);
+          bf8e_fma_pair(ta, e8, xp[r0], xm[r0], accp, accm);
+          fp8x4_decode(e16.z, e8);
+          fp8x4_decode(e16.w, e8 + 4);
+          bf8e_fma_pair(tb_, e8, xp[r0 + 1], xm[r0 + 1], accp, accm);
+        };
+        int i2 = ip;
+        const int step2 = PART * 2;
+        if (i2 + PART < I2) {
+          // 2-wide (4 rows) software-pipelined double buffer, mirroring the
+          // bf16 path's issue batching at the same register footprint
+          uint4 e0 = lde8(i2), e1 = lde8(i2 + PART);
+          uint4 t00 = ldt(2 * i2), t01 = ldt(2 * i2 + 1);
+          uint4 t10 = ldt(2 * (i2 + PART)), t11 = ldt(2 * (i2 + PART) + 1);
+          for (; i2 + 3 * PART < I2; i2 += step2) {
+            const uint4 ne0 = lde8(i2 + 2 * PART), ne1 = lde8(i2 + 3 * PART);
+            const uint4 nt00 = ldt(2 * (i2 + 2 * PART)),
+                        nt01 = ldt(2 * (i2 + 2 * PART) + 1),
+                        nt10 = ldt(2 * (i2 + 3 * PART)),
+                        nt11 = ldt(2 * (i2 + 3 * PART) + 1);
+            fma_pairblk(e0, t00, t01, 2 * i2);
+            fma_pairblk(e1, t10, t11, 2 * (i2 + PART));
+            e0 = ne0; e1 = ne1;
+            t00 = nt00; t01 = nt01; t10 = nt10; t11 = nt11;
+          }
+          fma_pairblk(e0, t00, t01, 2 * i2);
+          fma_pairblk(e1, t10, t11, 2 * (i2 + PART));
+          i2 += step2;
+        }
+        for (; i2 < I2; i2 += PART)
+          fma_pairblk(lde8(i2), ldt(2 * i2), ldt(2 * i2 + 1), 2 * i2);
+#pragma unroll
+        for (int q = 0; q < 8; ++q) {
+          partial[(ip * OCT + oi) * 8 + q] = accp[q];
+          partm[(ip * OCT + oi) * 8 + q] = accm[q];
+        }
+      }
+      __syncthreads();
+      for (int o = tid; o < O; o += nthreads) {
+        const float tB = bf2f(TBs[o]), eB = fp8_byte(EBs8[o]);
+        float sp = tB + eB, sm = tB - eB;
+        const int oo = o >> 3, j = o & 7;
+        for (int p = 0; p < PART; ++p) {
+          sp += partial[(p * OCT + oo) * 8 + j];
+          sm += partm[(p * OCT + oo) * 8 + j];
+        }
+        yp[o] = do_act ? tanhf(sp) : sp;
+        ym[o] = do_act ? tanhf(sm) : sm;
+      }
+    } else {
+      // scalar path (small / odd layers): plain element-ordered fp8
+      const int PART = nthreads / O;
+      if (PART > 1) {
+        const int oi = tid % O, ip = tid / O;
+        float accp = 0.0f, accm = 0.0f;
+        if (ip < PART) {
+          for (int i = ip; i < I; i += PART) {
+            const float tw = bf2f(Tt[(int64_t)i * O + oi]);
+            const float ew = fp8_byte(Et8[(int64_t)i * O + oi]);
+            accp = fmaf(tw + ew, xp[i], accp);
+            accm = fmaf(tw - ew, xm[i], accm);
+          }
+          partial[ip * O + oi] = accp;
+          partm[ip * O + oi] = accm;
+        }
+        __syncthreads();
+        for (int o = tid; o < O; o += nthreads) {
+          const float tB = bf2f(TBs[o]), eB = fp8_byte(EBs8[o]);
+          float sp = tB + eB, sm = tB - eB;
+          for (int p = 0; p < PART; ++p) {
+            sp += partial[p * O + o];
+            sm += partm[p * O + o];
+          }
+          yp[o] = do_act ? tanhf(sp) : sp;
+          ym[o] = do_act ? tanhf(sm) : sm;
+        }
+      } else {
+        for (int o = tid; o < O; o += nthreads) {
+          const float tB = bf2f(TBs[o]), eB = fp8_byte(EBs8[o]);
+          float accp = tB + eB, accm = tB - eB;
+          for (int i = 0; i < I; ++i) {
+            const float tw = bf2f(Tt[(int64_t)i * O + o]);
+            const float ew = fp8_byte(Et8[(int64_t)i * O + o]);
+            accp = fmaf(tw + ew, xp[i], accp);
+            accm = fmaf(tw - ew, xm[i], accm);
+          }
+          yp[o] = do_act ? tanhf(accp) : accp;
+          ym[o] = do_act ? tanhf(accm) : accm;
+        }
+      }
+    }
+    __syncthreads();
+    float* t = xp; xp = yp; yp = t;
+    t = xm; xm = ym; ym = t;
+  }
+  *xp_out = xp;
+  *xm_out = xm;
+}
+
 __device__ __forceinline__ float es_actnoise(uint64_t seed, uint64_t ctr) {
   esrng::f32x4 v = esrng::normal4(ctr, seed, 0xACu);
   return v.x;

@@ -9,6 +9,7 @@
 // permutation lives host-side in the engine. sign[b] = +1/-1 for antithetic
 // pairs and 0 for the noiseless-evaluation slot (then out = theta exactly).
 #include "common.h"
+#include "mlp_core.h"
 
 __global__ void pheno_bf16_kernel(uint16_t* __restrict__ out, const float* __restrict__ theta,
                                   const float* __restrict__ table,
@@ -44,6 +45,71 @@ __global__ void pheno_bf16_kernel(uint16_t* __restrict__ out, const float* __res
     *reinterpret_cast<u16x8*>(ob + t) = v;
   }
   for (; t < n_params; ++t) ob[t] = f2bf(theta[t] + s * noise[t]);
+}
+
+// ---- fp8 sigma*eps blob -----------------------------------------------------
+// The antithetic-pair rollout's dominant cost is streaming the per-pair
+// sigma*eps rows from HBM. fp8 (OCP e4m3fn) halves those bytes AND halves
+// the load count (one 16-B load covers 16 elements = one octet of TWO
+// consecutive i-rows), while the +-e antithetic cancellation stays exact
+// (both members decode the SAME quantized e). Weight blocks of vectorizable
+// layers are stored ROW-PAIR INTERLEAVED so the 16-B load is contiguous:
+//   byte(i, o) = woff + (i & ~1)*O + (o>>3)*16 + (i & 1)*8 + (o & 7)
+// scalar-path layers and biases stay element-ordered (1 byte per element).
+__device__ __forceinline__ int64_t fp8_src_elem(const MlpShape& sh, int64_t b) {
+  for (int l = 0; l < sh.n_layers; ++l) {
+    if (b < sh.boff[l]) {  // weight block l (b >= woff[l] by construction)
+      if (!sh.vec_ok[l] || (sh.dims[l] & 1)) return b;  // plain layout
+      const int O = sh.dims[l + 1];
+      const int64_t rel = b - sh.woff[l];
+      const int64_t pairblk = rel / (2 * O);
+      const int within = (int)(rel % (2 * O));
+      const int col = (within >> 4) * 8 + (within & 7);
+      const int half = (within >> 3) & 1;
+      return sh.woff[l] + (2 * pairblk + half) * (int64_t)O + col;
+    }
+    const int64_t bias_end = sh.boff[l] + sh.dims[l + 1];
+    if (b < bias_end) return b;  // bias: plain
+  }
+  return b;
+}
+
+__global__ void pheno_fp8_kernel(uint8_t* __restrict__ out, const float* __restrict__ table,
+                                 const int64_t* __restrict__ offsets, MlpShape sh,
+                                 int64_t n_params, int64_t row_stride, float std) {
+  const int64_t b = blockIdx.y;
+  const float* noise = table + offsets[b];
+  uint8_t* ob = out + b * row_stride;
+  // one thread per 4 output bytes -> one uint32 store
+  const int64_t stride4 = (int64_t)gridDim.x * blockDim.x * 4;
+  for (int64_t t = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * 4;
+       t < row_stride; t += stride4) {
+    float v[4];
+#pragma unroll
+    for (int k = 0; k < 4; ++k) {
+      const int64_t byte = t + k;
+      v[k] = 0.0f;
+      if (byte < n_params) v[k] = std * noise[fp8_src_elem(sh, byte)];
+    }
+    *reinterpret_cast<uint32_t*>(ob + t) = fp8x4_encode(v[0], v[1], v[2], v[3]);
+  }
+}
+
+extern "C" int es_pheno_fp8(void* out, const void* table, const void* offsets,
+                            const int32_t* dims_host, int32_t ndims, int64_t n_pop,
+                            int64_t n_params, int64_t row_stride, float std,
+                            void* stream) {
+  MlpShape sh;
+  int rc = mlp_shape_init(&sh, dims_host, ndims, row_stride);
+  if (rc) return rc;
+  int threads = 256;
+  int bx = (int)std::min<int64_t>((row_stride + threads * 4 - 1) / (threads * 4), 1024);
+  dim3 grid(bx, (unsigned)n_pop);
+  pheno_fp8_kernel<<<grid, dim3(threads), 0, (hipStream_t)stream>>>(
+      (uint8_t*)out, (const float*)table, (const int64_t*)offsets, sh, n_params,
+      row_stride, std);
+  ES_CHECK_LAUNCH();
+  return 0;
 }
 
 extern "C" int es_pheno_bf16(void* out, const void* theta, const void* table,

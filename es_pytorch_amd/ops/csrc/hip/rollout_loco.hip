@@ -460,9 +460,11 @@ __device__ __forceinline__ void loco_dyn_partials_pair(
   __syncthreads();
 }
 
+// EB = uint16_t (bf16 eps rows) or uint8_t (fp8 row-pair-interleaved rows)
+template <typename EB>
 __device__ __forceinline__ void loco_pair_step_body(
     const MlpShape& sh, const LocoArgs& la, const LocoPtrs& P,
-    const uint16_t* tb, const uint16_t* eb, int bp, int bm, uint64_t salt,
+    const uint16_t* tb, const EB* eb, int bp, int bm, uint64_t salt,
     float* bufAp, float* bufAm, float* bufBp, float* bufBm, float* partial,
     float* rawsP, float* rawsM, float* abufP, float* abufM) {
   const int tid = threadIdx.x, nth = blockDim.x;
@@ -470,8 +472,12 @@ __device__ __forceinline__ void loco_pair_step_body(
   loco_build_obs(la, P, bm, bufAm, rawsM, tid, nth);
   __syncthreads();
   float *ap, *am;
-  mlp_layers_pair(tb, eb, sh, bufAp, bufAm, bufBp, bufBm, partial, tid, nth, 1,
-                  &ap, &am);
+  if constexpr (sizeof(EB) == 1)
+    mlp_layers_pair_fp8(tb, (const uint8_t*)eb, sh, bufAp, bufAm, bufBp, bufBm,
+                        partial, tid, nth, 1, &ap, &am);
+  else
+    mlp_layers_pair(tb, (const uint16_t*)eb, sh, bufAp, bufAm, bufBp, bufBm,
+                    partial, tid, nth, 1, &ap, &am);
   loco_decode_action(sh, la, P, bp, salt, ap, abufP, tid);
   loco_decode_action(sh, la, P, bm, salt, am, abufM, tid);
   __syncthreads();
@@ -497,7 +503,7 @@ __device__ __forceinline__ void loco_pair_step_body(
   const int p = q / la.eps, e = q % la.eps;                      \
   const int bp = p * la.eps + e;                                 \
   const int bm = (n_pairs + p) * la.eps + e;                     \
-  const uint16_t* ebp = eb + (int64_t)p * la.row_stride;
+  const auto* ebp = eb + (int64_t)p * la.row_stride;
 
 // ES_PAIR_MINWAVES (experiment knob): force a min-waves/SIMD bound on the
 // pair step kernel so deeper rings can be capped to 3 waves (168 VGPRs)
@@ -507,6 +513,14 @@ __device__ __forceinline__ void loco_pair_step_body(
 __global__ void __launch_bounds__(256, ES_PAIR_MINWAVES)
 loco_pair_step_kernel(MlpShape sh, LocoArgs la, LocoPtrs P, const uint16_t* tb,
                       const uint16_t* eb, int n_pairs, uint64_t salt) {
+  ES_LOCO_PAIR_CARVE();
+  loco_pair_step_body(sh, la, P, tb, ebp, bp, bm, salt, bufAp, bufAm, bufBp, bufBm,
+                      partial, rawsP, rawsM, abufP, abufM);
+}
+
+__global__ void __launch_bounds__(256, ES_PAIR_MINWAVES)
+loco_pair_step_fp8_kernel(MlpShape sh, LocoArgs la, LocoPtrs P, const uint16_t* tb,
+                          const uint8_t* eb, int n_pairs, uint64_t salt) {
   ES_LOCO_PAIR_CARVE();
   loco_pair_step_body(sh, la, P, tb, ebp, bp, bm, salt, bufAp, bufAm, bufBp, bufBm,
                       partial, rawsP, rawsM, abufP, abufM);
@@ -684,6 +698,39 @@ extern "C" int es_loco_pair_step(
   const unsigned grid = (unsigned)(n_pairs * la.eps);
   loco_pair_step_kernel<<<dim3(grid), dim3(256), lds, (hipStream_t)stream>>>(
       sh, la, P, (const uint16_t*)theta_row, (const uint16_t*)eps_rows, n_pairs, salt);
+  ES_CHECK_LAUNCH();
+  return 0;
+}
+
+// fp8-eps pair step: eps_rows = (n_pairs, row_stride) BYTES, written by
+// es_pheno_fp8 (row-pair-interleaved e4m3); theta_row stays bf16.
+extern "C" int es_loco_pair_step_fp8(
+    const void* theta_row, const void* eps_rows, const void* obmean,
+    const void* obstd, const int32_t* dims_host, int32_t ndims, const void* seed_dev,
+    uint64_t salt, float ob_clip, const void* ac_std_dev, int64_t row_stride,
+    void* s_glob, void* pos, const void* goal, const void* Am, const void* Bm,
+    const void* b0, const void* wv, const void* wa, const void* wy, const void* wh,
+    void* alive, void* rew_total, void* member_steps, void* behv, void* mo_sum,
+    void* mo_sumsq, int32_t n_pairs, int32_t sdim, int32_t adim, int32_t goal_flag,
+    int32_t terminate, int32_t noiseless_from, int32_t bins, int32_t eps,
+    int32_t act_mode, float leak, float ctrl, float alive_bonus, float fall_thr,
+    float dt, void* stream) {
+  MlpShape sh;
+  LocoArgs la;
+  unsigned lds_unused;
+  int rc = loco_prepare(&sh, &la, dims_host, ndims, row_stride, ob_clip, sdim, adim,
+                        goal_flag, terminate, noiseless_from, bins, eps, act_mode, leak,
+                        ctrl, alive_bonus, fall_thr, dt, &lds_unused);
+  if (rc) return rc;
+  LocoPtrs P = loco_ptrs(nullptr, obmean, obstd, ac_std_dev, seed_dev, s_glob, pos,
+                         goal, Am, Bm, b0, wv, wa, wy, wh, alive, rew_total,
+                         member_steps, behv, mo_sum, mo_sumsq);
+  const int Spad = (sdim + 3) & ~3;
+  const unsigned lds =
+      (unsigned)((4 * sh.maxdim + 2 * 256 * 8 + 2 * Spad + 2 * 64) * sizeof(float));
+  const unsigned grid = (unsigned)(n_pairs * la.eps);
+  loco_pair_step_fp8_kernel<<<dim3(grid), dim3(256), lds, (hipStream_t)stream>>>(
+      sh, la, P, (const uint16_t*)theta_row, (const uint8_t*)eps_rows, n_pairs, salt);
   ES_CHECK_LAUNCH();
   return 0;
 }

@@ -31,6 +31,7 @@ def main():
     p.add_argument("--pair", action="store_true", help="antithetic-pair rollout")
     p.add_argument("--dyn-group", type=int, default=5,
                    help="members per dynamics block in split mode")
+    p.add_argument("--fp8", action="store_true", help="fp8 sigma*eps stream")
     args = p.parse_args()
 
     from es_pytorch_amd.config import AttrDict
@@ -63,7 +64,8 @@ def main():
     rs = np.random.RandomState(0)
     eng = GpuEngine(cfg, comm, policy, nt, env, rs, use_graph=args.graph,
                     rollout_mode="episode" if args.episode else "step",
-                    split_dyn=args.split or None, pair_rollout=args.pair or None)
+                    split_dyn=args.split or None, pair_rollout=args.pair or None,
+                    eps_fp8=args.fp8 or None)
     ranker = CenteredRanker()
     eng.step(ranker)  # warmup
 
