@@ -290,6 +290,13 @@ class GpuEngine:
             off += I * O + O
         return True
 
+    def _grad_qstd(self) -> float:
+        """Non-zero => the gather re-quantizes each noise value through the
+        e4m3 round trip at sigma, so the update uses EXACTLY the perturbation
+        values the fp8 rollout evaluated (estimator-exact ES on the quantized
+        perturbation distribution)."""
+        return float(self.policy.std) if self.eps_fp8 else 0.0
+
     # ------------------------------------------------------------------ ops
     def _stream(self):
         return torch.cuda.current_stream(self.device).cuda_stream if \
@@ -696,7 +703,8 @@ class GpuEngine:
             ri = all_rows_dev[:, 2].long().contiguous()
             ops.check(ops.hip().es_grad_gather(
                 self.grad.data_ptr(), self.nt.noise.data_ptr(), rf.data_ptr(),
-                ri.data_ptr(), rf.numel(), self.n, self._stream()), "es_grad_gather")
+                ri.data_ptr(), rf.numel(), self.n, self._grad_qstd(),
+                self._stream()), "es_grad_gather")
             self._optim_step(float(npop), float(cfg.policy.l2coeff))
             # host mirror (for reporters and the entry scripts' heuristics)
             # overlaps the update kernels above
@@ -723,7 +731,8 @@ class GpuEngine:
                                                        dtype=np.int64)).to(self.device)
             ops.check(ops.hip().es_grad_gather(
                 self.grad.data_ptr(), self.nt.noise.data_ptr(), rf.data_ptr(),
-                ri.data_ptr(), rf.numel(), self.n, self._stream()), "es_grad_gather")
+                ri.data_ptr(), rf.numel(), self.n, self._grad_qstd(),
+                self._stream()), "es_grad_gather")
             self._optim_step(float(ranker.n_fits_ranked), float(cfg.policy.l2coeff))
         self.sync_host(light=True)
         pos, neg = all_rows[:, :O], all_rows[:, O:2 * O]

@@ -180,7 +180,7 @@ class MultiAgentGpuEngine:
                 rankers[i].noise_inds, dtype=np.int64)).to(self.device)
             ops.check(ops.hip().es_grad_gather(
                 st.grad.data_ptr(), self.nt.noise.data_ptr(), rf.data_ptr(),
-                ri.data_ptr(), rf.numel(), st.n, self._stream()), "es_grad_gather")
+                ri.data_ptr(), rf.numel(), st.n, 0.0, self._stream()), "es_grad_gather")
             opt = st.policy.optim
             assert isinstance(opt, Adam)
             opt.t += 1

@@ -170,7 +170,7 @@ def _bind_hip(lib):
     lib.es_pheno_fp8.argtypes = [p, p, p, p, i32, i64, i64, i64, f32, p]
     lib.es_mlp_fwd.argtypes = [p, p, p, p, p, p, i32, p, u64, i32, f32, p, i64, i32,
                                i32, i32, i32, i32, p, p, p]
-    lib.es_grad_gather.argtypes = [p, p, p, p, i64, i64, p]
+    lib.es_grad_gather.argtypes = [p, p, p, p, i64, i64, f32, p]
     lib.es_adam_step.argtypes = [p, p, p, p, i64, f32, f32, f32, f32, f32, f32, p]
     lib.es_sgd_step.argtypes = [p, p, p, i64, f32, f32, f32, f32, p]
     lib.es_loco_step.argtypes = [p, p, p, p, i32, p, u64, f32, p, i64,

@@ -58,3 +58,11 @@ __device__ __forceinline__ float fp8_byte(uint8_t b) {
   es_f32x2 lo = __builtin_amdgcn_cvt_pk_f32_fp8((uint32_t)b, false);
   return lo.x;
 }
+
+// e4m3fn quantization round trip through the same hardware converters the
+// pheno/rollout path uses — yields EXACTLY the value the rollout computed
+__device__ __forceinline__ float e4m3_roundtrip(float v) {
+  const uint32_t w = __builtin_amdgcn_cvt_pk_fp8_f32(v, v, 0u, false);
+  es_f32x2 lo = __builtin_amdgcn_cvt_pk_f32_fp8(w, false);
+  return lo.x;
+}

@@ -12,11 +12,17 @@
 
 #define ES_GRAD_TILE 256
 
-template <int ES_GRAD_ELEMS>
+// Q: re-quantize each table value through the e4m3 round trip at sigma
+// `qstd` before the FMA — the fp8-eps rollout evaluated fitness at
+// perturbations q(sigma*eps)/sigma, and an ES estimator is exact when the
+// gather uses the SAME perturbation values it evaluated (g = sum f_p * e'_p).
+template <int ES_GRAD_ELEMS, bool Q>
 __global__ void __launch_bounds__(256)
 grad_gather_kernel(float* __restrict__ g, const float* __restrict__ table,
                    const float* __restrict__ fits, const int64_t* __restrict__ offsets,
-                   int64_t n_pop, int64_t n_params) {
+                   int64_t n_pop, int64_t n_params, float qstd) {
+  const float qinv = Q ? 1.0f / qstd : 0.0f;
+  auto xform = [&](float x) { return Q ? e4m3_roundtrip(qstd * x) * qinv : x; };
   __shared__ float s_fit[ES_GRAD_TILE];
   __shared__ int64_t s_off[ES_GRAD_TILE];
 
@@ -43,13 +49,13 @@ grad_gather_kernel(float* __restrict__ g, const float* __restrict__ table,
       const float* row0 = table + s_off[0] + base;
 #pragma unroll
       for (int k = 0; k < ES_GRAD_ELEMS; ++k)
-        c[k] = row0[k * blockDim.x + threadIdx.x];
+        c[k] = xform(row0[k * blockDim.x + threadIdx.x]);
       for (int i = 0; i < tile - 1; ++i) {
         const float* rown = table + s_off[i + 1] + base;
         float nx[ES_GRAD_ELEMS];
 #pragma unroll
         for (int k = 0; k < ES_GRAD_ELEMS; ++k)
-          nx[k] = rown[k * blockDim.x + threadIdx.x];
+          nx[k] = xform(rown[k * blockDim.x + threadIdx.x]);
         const float f = s_fit[i];
 #pragma unroll
         for (int k = 0; k < ES_GRAD_ELEMS; ++k) {
@@ -67,7 +73,8 @@ grad_gather_kernel(float* __restrict__ g, const float* __restrict__ table,
 #pragma unroll
         for (int k = 0; k < ES_GRAD_ELEMS; ++k) {
           const int64_t t = base + k * blockDim.x + threadIdx.x;
-          if (t < n_params) acc[k] = fmaf(f, row[k * blockDim.x + threadIdx.x], acc[k]);
+          if (t < n_params)
+            acc[k] = fmaf(f, xform(row[k * blockDim.x + threadIdx.x]), acc[k]);
         }
       }
     }
@@ -82,7 +89,7 @@ grad_gather_kernel(float* __restrict__ g, const float* __restrict__ table,
 
 extern "C" int es_grad_gather(void* g, const void* table, const void* fits,
                               const void* offsets, int64_t n_pop, int64_t n_params,
-                              void* stream) {
+                              float qstd, void* stream) {
   const int threads = 256;
   // pick the per-thread element count so the grid fills the chip (a fixed
   // ELEMS=4 gave only 163 blocks at n=167k — occupancy-starved)
@@ -92,18 +99,16 @@ extern "C" int es_grad_gather(void* g, const void* table, const void* fits,
     elems >>= 1;
   const int64_t chunk = (int64_t)threads * elems;
   const int blocks = (int)((n_params + chunk - 1) / chunk);
-  if (elems == 4)
-    grad_gather_kernel<4><<<dim3(blocks), dim3(threads), 0, (hipStream_t)stream>>>(
-        (float*)g, (const float*)table, (const float*)fits, (const int64_t*)offsets,
-        n_pop, n_params);
-  else if (elems == 2)
-    grad_gather_kernel<2><<<dim3(blocks), dim3(threads), 0, (hipStream_t)stream>>>(
-        (float*)g, (const float*)table, (const float*)fits, (const int64_t*)offsets,
-        n_pop, n_params);
-  else
-    grad_gather_kernel<1><<<dim3(blocks), dim3(threads), 0, (hipStream_t)stream>>>(
-        (float*)g, (const float*)table, (const float*)fits, (const int64_t*)offsets,
-        n_pop, n_params);
+  const bool q = qstd > 0.0f;
+#define ES_LAUNCH_GG(E, QF)                                                       \
+  grad_gather_kernel<E, QF><<<dim3(blocks), dim3(threads), 0,                     \
+                              (hipStream_t)stream>>>(                             \
+      (float*)g, (const float*)table, (const float*)fits,                         \
+      (const int64_t*)offsets, n_pop, n_params, qstd)
+  if (elems == 4) { if (q) ES_LAUNCH_GG(4, true); else ES_LAUNCH_GG(4, false); }
+  else if (elems == 2) { if (q) ES_LAUNCH_GG(2, true); else ES_LAUNCH_GG(2, false); }
+  else { if (q) ES_LAUNCH_GG(1, true); else ES_LAUNCH_GG(1, false); }
+#undef ES_LAUNCH_GG
   ES_CHECK_LAUNCH();
   return 0;
 }

@@ -51,7 +51,7 @@ def scale_noise(fits: np.ndarray, noise_inds: np.ndarray, nt: NoiseTable, policy
         stream = torch.cuda.current_stream(dev).cuda_stream
         ops.check(ops.hip().es_grad_gather(g.data_ptr(), nt.noise.data_ptr(),
                                            fits_t.data_ptr(), offs_t.data_ptr(),
-                                           len(fits), policy_len, stream),
+                                           len(fits), policy_len, 0.0, stream),
                   "es_grad_gather")
         return g
 

@@ -114,6 +114,33 @@ def test_fp8_sigma0_matches_bf16_pair(dev):
     np.testing.assert_allclose(outs[True], outs[False], rtol=2e-4, atol=2e-3)
 
 
+def test_quantized_gather_matches_rollout_perturbation(dev):
+    """es_grad_gather(qstd=sigma) must use EXACTLY the perturbation values
+    the fp8 rollout evaluated: with a single unit fitness, sigma * g equals
+    the decoded pheno_fp8 blob elementwise (same hardware converters on
+    both paths -> estimator-exact ES on the quantized distribution)."""
+    from es_pytorch_amd import ops
+    eng = _mk(dev, fp8=True)
+    eng._upload_offsets()
+    eng._pheno()
+    torch.cuda.synchronize(dev)
+    std = float(eng.policy.std)
+
+    fits = torch.ones(1, dtype=torch.float32, device=dev)
+    offs = eng.offsets[:1].contiguous()
+    g = torch.empty(eng.n, dtype=torch.float32, device=dev)
+    stream = torch.cuda.current_stream(dev).cuda_stream
+    ops.check(ops.hip().es_grad_gather(g.data_ptr(), eng.nt.noise.data_ptr(),
+                                       fits.data_ptr(), offs.data_ptr(), 1, eng.n,
+                                       std, stream), "grad_q")
+    torch.cuda.synchronize(dev)
+
+    blob = eng.eps_rows[0].cpu().numpy()
+    decoded = e4m3fn_decode(blob[_interleave_map(eng.dims, eng.n)]).astype(np.float64)
+    np.testing.assert_allclose(std * g.cpu().numpy().astype(np.float64), decoded,
+                               rtol=1e-6, atol=1e-12)
+
+
 def test_fp8_fidelity_at_flagship_sigma(dev):
     """sigma=0.02, 200 steps, pop 512: member fitness ranking and the
     reconstructed gradient must track the bf16 pair path closely."""

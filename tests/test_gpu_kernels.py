@@ -161,7 +161,7 @@ def test_grad_gather_kernel(dev):
     fits = torch.randn(P, device=dev)
     g = torch.empty(n, device=dev)
     ops.check(ops.hip().es_grad_gather(g.data_ptr(), table.data_ptr(), fits.data_ptr(),
-                                       offs.data_ptr(), P, n, _stream(dev)), "grad")
+                                       offs.data_ptr(), P, n, 0.0, _stream(dev)), "grad")
     torch.cuda.synchronize()
     rows = torch.stack([table[o:o + n] for o in offs.cpu()])
     expect = fits.cpu() @ rows.cpu()
