@@ -49,10 +49,14 @@ def main():
                         "measured 1.27x over per-member weight blobs, and "
                         "the perturbation keeps MORE bf16 mantissa)")
     p.add_argument("--cpu", action="store_true", help="debug: run the engine on CPU")
-    p.add_argument("--eps-fp8", action="store_true",
-                   help="fp8 (e4m3) sigma*eps stream: halves the dominant "
-                        "HBM weight bytes; fitness fidelity documented in "
-                        "profiles/pair_fidelity_sweep.txt")
+    p.add_argument("--no-fp8", action="store_true",
+                   help="disable the fp8 (e4m3) sigma*eps stream (the "
+                        "flagship default: halves the dominant HBM weight "
+                        "bytes AND load count; the gradient gather re-"
+                        "quantizes through the same e4m3 hardware so the "
+                        "update is estimator-exact ES on the quantized "
+                        "perturbation distribution; fidelity + learning "
+                        "evidence in profiles/)")
     p.add_argument("--mode", type=str, default="step", choices=["step", "episode"],
                    help="rollout launch shape: per-step graph-replayed grid, or "
                         "one whole-episode launch per generation (blocks drift; "
@@ -117,7 +121,7 @@ def main():
         engine = GpuEngine(cfg, comm, policy, nt, env, rs, objective=args.objective,
                            use_graph=not args.no_graph, rollout_mode=args.mode,
                            pair_rollout=False if args.no_pair else None,
-                           eps_fp8=args.eps_fp8 or None)
+                           eps_fp8=not (args.no_fp8 or args.no_pair))
         if args.objective == "nsr":
             # seeded starter archive on device (NSR-A semantics: novelty vs
             # the behaviour archive, grown per generation)
@@ -187,6 +191,8 @@ def main():
                 "objective": args.objective,
                 "parallelism": f"dp{args.gpus}",
                 "pair_rollout": bool(getattr(engine, "pair_rollout", False)),
+                "eps_encoding": ("e4m3" if getattr(engine, "eps_fp8", False)
+                                 else "bf16"),
                 "gens_per_sec": round(gens_per_sec, 3),
                 "noise_table_elems": tbl,
                 "n_params": len(policy),
