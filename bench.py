@@ -146,9 +146,9 @@ def main():
     for _ in range(args.steps):
         tr, _ = engine.step(ranker) if use_cuda else (engine.step(ranker), None)
         if args.objective == "nsr" and use_cuda:
-            b = comm.broadcast_obj(list(tr.behaviour), src=0)
-            engine.archive = torch.cat([engine.archive, torch.tensor(
-                [b], dtype=torch.float64, device=device)])
+            # on-device archive growth inside the timed region (reference
+            # nsra.py:130-135): ONE 2-float broadcast + device cdist re-score
+            engine.grow_archive()
         steps_done += engine.timings["env_steps"]
     comm.barrier()
     if use_cuda:

@@ -765,6 +765,21 @@ class GpuEngine:
                         "env_steps": steps}
         return noiseless, gen_obstat
 
+    def grow_archive(self) -> float:
+        """Device-resident NSR-A archive growth (reference ``nsra.py:130-135``
+        + ``novelty.py:9-13``): append rank-0's noiseless behaviour to the
+        device archive (ONE 2-float RCCL broadcast — the entry must be
+        identical on every rank) and return its novelty against the archive
+        it joined. Call once per generation after :meth:`step`; stays on
+        device end to end."""
+        assert self.archive is not None, "set engine.archive first"
+        b = self._member_behv()[-1][:2].double().contiguous()
+        self.comm.broadcast_tensor_(b, src=0)
+        nov = float(novelty_batch(b.unsqueeze(0), self.archive,
+                                  self.novelty_k)[0].item())
+        self.archive = torch.cat([self.archive, b.unsqueeze(0)])
+        return nov
+
     def _upload_offsets(self):
         """Sample this rank's antithetic noise offsets and upload to device."""
         offs = self.nt.sample_idxs(self.rs, self.pairs)

@@ -86,6 +86,17 @@ class Comm:
         self.allreduce_sum_(t)
         return t.item()
 
+    def broadcast_tensor_(self, t: torch.Tensor, src: int = 0) -> torch.Tensor:
+        """In-place tensor broadcast (device buffers under nccl=RCCL)."""
+        if not self.initialized or self.size == 1:
+            return t
+        moved = t.device != self._coll_device
+        buf = t.to(self._coll_device) if moved else t
+        dist.broadcast(buf, src=src)
+        if moved:
+            t.copy_(buf.to(t.device))
+        return t
+
     def broadcast_obj(self, obj: Any, src: int = 0) -> Any:
         """Broadcast a picklable object (reference ``comm.scatter([x]*size)`` idiom)."""
         if not self.initialized or self.size == 1:

@@ -170,7 +170,11 @@ def main(cfg):
             eng = engines[idx]
             eng.archive = torch.from_numpy(np.asarray(archive)).to(device)
             tr, gen_obstat = eng.step(ranker, reporter)
-            behv = comm.broadcast_obj(list(tr.behaviour), src=0)
+            # archive growth + novelty re-score on device (ONE 2-float
+            # broadcast); host mirror kept for checkpointing/saving
+            nov = eng.grow_archive()
+            archive = eng.archive.cpu().numpy()
+            behv = list(archive[-1])
         else:
             tr, gen_obstat = es.step(cfg, comm, population[idx], nt, env, ns_fn, rs,
                                      ranker, reporter)
@@ -184,8 +188,10 @@ def main(cfg):
             if use_gpu:
                 engines[i]._push_obstat()
 
-        nov = comm.broadcast_obj(novelty(np.asarray(behv), archive, cfg.novelty.k), src=0)
-        archive = update_archive(comm, behv, archive)
+        if not use_gpu:
+            nov = comm.broadcast_obj(novelty(np.asarray(behv), archive, cfg.novelty.k),
+                                     src=0)
+            archive = update_archive(comm, behv, archive)
         archive_box["archive"] = archive
         policies_novelties[idx] = nov
 

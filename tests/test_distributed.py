@@ -110,3 +110,18 @@ def _es_e2e_worker(rank, world):
 @pytest.mark.timeout(300)
 def test_es_end_to_end_mp():
     assert all(run_mp(_es_e2e_worker, world=2, timeout=280))
+
+
+def _broadcast_tensor_worker(rank, world):
+    from es_pytorch_amd.parallel.comm import Comm
+    comm = Comm(torch.device("cpu"))
+    # rank-dependent input, rank-0's value expected everywhere
+    t = torch.tensor([float(rank + 1), float(10 * rank)], dtype=torch.float64)
+    comm.broadcast_tensor_(t, src=0)
+    np.testing.assert_allclose(t.numpy(), [1.0, 0.0])
+    return True
+
+
+def test_broadcast_tensor_mp():
+    """Comm.broadcast_tensor_ (the archive-growth primitive) across 2 ranks."""
+    assert all(run_mp(_broadcast_tensor_worker, world=2))
