@@ -17,6 +17,13 @@ update (reference README.md:10-12).
 This module is the EPISODIC path (works on CPU, any env, any fit_fn). The
 GPU-batched whole-generation path with the same semantics lives in
 ``core/engine.py``.
+
+
+PROVENANCE: the step/test_params/approx_grad signatures and the generation
+control flow are deliberately ported from the reference (src/core/es.py) —
+they ARE the public API contract this framework preserves; the collective
+layer underneath (parallel/comm.py) and the GPU whole-generation engine
+(core/engine.py) are original MI355X-native code.
 """
 from __future__ import annotations
 

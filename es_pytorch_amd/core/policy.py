@@ -11,6 +11,13 @@ and optimizer state.
 The GPU engine (``core/engine.py``) treats this object as the host-side truth
 it syncs with once per generation; perturbation there is the batched HIP
 pheno kernel, not this per-module path.
+
+
+PROVENANCE: the attribute layout, flat-vector semantics and pickle format
+are pinned by the reference checkpoint contract (src/core/policy.py) and
+parts of this file are a direct port of those definitions; additions
+(tensor-noise handling, compat unpickler for reference checkpoints, safe
+file handling) are original.
 """
 from __future__ import annotations
 

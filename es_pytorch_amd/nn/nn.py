@@ -11,6 +11,12 @@ These torch modules are the EPISODIC/CPU path and the checkpoint phenotype.
 The GPU training path evaluates the same architecture population-batched via
 the fused HIP MLP kernel (``ops/csrc/hip/mlp_fwd.hip``); layer sizes and
 normalization semantics are shared through :meth:`BaseNet.layer_dims`.
+
+
+PROVENANCE: the module structure and forward bodies follow the reference
+(src/nn/nn.py) closely — checkpoint compatibility pins the state_dict
+layout; layer_dims(), _normalize(), the unified _ActionView output sizing
+and the kernel-shared contracts are original.
 """
 from __future__ import annotations
 

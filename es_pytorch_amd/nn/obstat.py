@@ -8,6 +8,11 @@ stat at eps=1e-2 (``src/core/policy.py:27``).
 The cross-rank merge replaces the reference's custom pickling MPI reduce op
 (``obstat.py:5-10,39-43``) with ONE packed fp64 all_reduce over
 ``[sum(ob_dim), sumsq(ob_dim), count]`` — RCCL-friendly, no pickling.
+
+
+PROVENANCE: the running-stat math (1e-2 std floor, eps seeding semantics)
+is the reference's (src/nn/obstat.py); the packed-tensor all_reduce merge
+replacing the pickling MPI op is original.
 """
 from __future__ import annotations
 

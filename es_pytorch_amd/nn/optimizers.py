@@ -15,6 +15,12 @@ The CPU implementation below is numpy (matching reference numerics); the GPU
 training engine uses the fused HIP Adam kernel (``ops/csrc/hip/update.hip``)
 which implements the identical formula on device — parity-tested in
 ``tests/test_gpu_kernels.py``.
+
+
+PROVENANCE: formulas ported verbatim from the reference (src/nn/optimizers
+.py, itself adapted from uber-research/deep-neuroevolution) — the fused
+HIP update kernel must match them bit-for-bit, so the host definitions
+cannot drift.
 """
 from __future__ import annotations
 

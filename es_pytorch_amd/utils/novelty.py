@@ -8,6 +8,10 @@ k nearest archive entries; new entries are produced on rank 0 and broadcast
 ``novelty_batch`` is the population-engine version: novelty of B behaviours
 against the archive in one torch cdist + topk (on device when the archive
 tensor lives there) — SURVEY.md K8.
+
+
+PROVENANCE: novelty/update_archive reproduce the reference math
+(src/utils/novelty.py); the batched on-device variants are original.
 """
 from __future__ import annotations
 

@@ -9,6 +9,12 @@ standard argsort-rank scaled to [-0.5, 0.5] (``rankers.py:53-58``).
 Shapes are tiny (population-sized), so this runs in numpy on every rank
 redundantly — identical inputs give identical updates, preserving the
 reference's no-parameter-broadcast design.
+
+
+PROVENANCE: the rank transforms are numeric definitions taken from the
+reference (src/utils/rankers.py) — OpenAI-ES centered rank, antithetic
+post-rank difference, elite re-indexing, weighted multi-objective; the
+math IS the spec, so the bodies match it closely by intent.
 """
 from __future__ import annotations
 

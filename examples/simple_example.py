@@ -10,6 +10,10 @@ Run (N CPU ranks, gloo):    torchrun --standalone --local-addr 127.0.0.1 \
 The inline generation block below is what ``es.step`` packages (reference
 ``simple_example.py:45-59``): test_params -> update obstat -> rank ->
 approx_grad, with fitness triples exchanged over gloo/RCCL.
+
+PROVENANCE: deliberately mirrors the reference's didactic script
+(simple_example.py) line for line where the inline generation loop is the
+point being demonstrated.
 """
 import os
 import sys
