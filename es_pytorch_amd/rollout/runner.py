@@ -94,7 +94,22 @@ def multi_agent_runner(policies: List[torch.nn.Module], env, max_steps: int,
                 break
 
     if not saved_obs:
-        saved_obs.append(np.zeros(np.asarray(obs).shape))
+        saved_obs.append([np.zeros(np.shape(o)) for o in obs])
 
     behv += behv[-3:] * (max_steps - int(len(behv) / 3))
-    return rews, behv, np.array(saved_obs), step
+    return rews, behv, _stack_agent_obs(saved_obs), step
+
+
+def _stack_agent_obs(saved_obs):
+    """(steps, agents, ob_dim) array; heterogeneous teams (per-agent obs of
+    different widths, e.g. a real Unity build with several behaviors) fall
+    back to a (steps, agents) object array — MultiAgentTrainingResult's
+    per-agent column reductions work on either."""
+    try:
+        return np.array(saved_obs)
+    except ValueError:
+        arr = np.empty((len(saved_obs), len(saved_obs[0])), dtype=object)
+        for i, step_obs in enumerate(saved_obs):
+            for j, o in enumerate(step_obs):
+                arr[i, j] = np.asarray(o)
+        return arr

@@ -56,3 +56,52 @@ class Discrete:
 
     def __repr__(self):
         return f"Discrete({self.n})"
+
+
+class MultiDiscrete:
+    def __init__(self, nvec):
+        self.nvec = np.asarray(nvec, dtype=np.int64)
+        self.shape = self.nvec.shape
+        self.dtype = np.int64
+        self._rs = np.random.RandomState()
+
+    def seed(self, seed=None):
+        self._rs = np.random.RandomState(seed)
+
+    def sample(self):
+        return (self._rs.random_sample(self.nvec.shape) * self.nvec).astype(np.int64)
+
+    def contains(self, x):
+        x = np.asarray(x)
+        return x.shape == self.shape and bool(np.all(x >= 0) and np.all(x < self.nvec))
+
+    def __repr__(self):
+        return f"MultiDiscrete({self.nvec.tolist()})"
+
+
+class Tuple:
+    """A fixed tuple of component spaces (multi-agent obs/action surfaces)."""
+
+    def __init__(self, spaces):
+        self.spaces = tuple(spaces)
+        self.shape = None
+
+    def seed(self, seed=None):
+        for i, s in enumerate(self.spaces):
+            s.seed(None if seed is None else seed + i)
+
+    def sample(self):
+        return tuple(s.sample() for s in self.spaces)
+
+    def contains(self, x):
+        return len(x) == len(self.spaces) and all(s.contains(v)
+                                                  for s, v in zip(self.spaces, x))
+
+    def __len__(self):
+        return len(self.spaces)
+
+    def __getitem__(self, i):
+        return self.spaces[i]
+
+    def __repr__(self):
+        return f"Tuple({list(self.spaces)})"
