@@ -236,14 +236,22 @@ class GpuEngine:
         # constructor True FORCES pair (tests, explicit intent); the config
         # flag is auto-gated on grid size: pair halves the block count, so
         # tiny populations would idle CUs (Hopper pop 256 -> 128 pair blocks
-        # measured 9.0M vs 12.4M env-steps/s fused); measured crossover is
-        # ~384 pair blocks (1.5/CU): pop 512 fused 58.0 vs pair 58.9 us/step,
-        # pop 768 73.3 vs 70.3, pop 1024 89.6 vs 75.2
+        # measured 9.0M vs 12.4M env-steps/s fused bf16). Crossovers are
+        # measured per eps encoding: bf16 pair wins from ~384 pair blocks
+        # (pop 512 fused 58.0 vs pair 58.9 us/step, 768 73.3 vs 70.3, 1024
+        # 89.6 vs 75.2); fp8 pair from ~256 blocks (round 2, same-box: pop
+        # 512 fused 53.5 vs pair-fp8 50.1, 640 61.2 vs 54.7, 768 69.0 vs
+        # 56.3, 1024 84.4 vs 59.1 — the halved load count pays earlier).
+        fp8_possible = (bool(cfg.general.get("eps_fp8", False))
+                        if eps_fp8 is None else bool(eps_fp8)) and \
+            self.rollout_mode == "step" and self.steps_per_launch == 1 and \
+            self._fp8_layout_ok(self.dims)
         if pair_rollout is not None:
             want_pair = bool(pair_rollout)
         else:
+            gate_blocks = 256 if fp8_possible else 384
             want_pair = (bool(cfg.general.get("pair_rollout", False))
-                         and self.pairs * self.eps >= 384)
+                         and self.pairs * self.eps >= gate_blocks)
         self.pair_rollout = (want_pair and self.fused and not self.split_dyn
                              and self.pairs >= 1)
         self._zero_off = torch.zeros(1, dtype=torch.int64, device=d)
@@ -254,10 +262,7 @@ class GpuEngine:
         # (even input dim + 16-B-aligned weight offsets on vec layers)
         want_fp8 = (bool(cfg.general.get("eps_fp8", False))
                     if eps_fp8 is None else bool(eps_fp8))
-        self.eps_fp8 = (want_fp8 and self.pair_rollout
-                        and self.rollout_mode == "step"
-                        and self.steps_per_launch == 1
-                        and self._fp8_layout_ok(self.dims))
+        self.eps_fp8 = fp8_possible and self.pair_rollout
         if want_fp8 and not self.eps_fp8 and comm.rank == 0:
             import sys
             print("[engine] eps_fp8 requested but unavailable for this "
