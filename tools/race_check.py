@@ -19,7 +19,7 @@ import torch
 sys.path.insert(0, os.path.join(os.path.dirname(os.path.abspath(__file__)), ".."))
 
 
-def run_once(gens, graph, fused):
+def run_once(gens, graph, fused, pair=None, fp8=None):
     from es_pytorch_amd.config import AttrDict
     from es_pytorch_amd.core.engine import GpuEngine
     from es_pytorch_amd.core.noisetable import NoiseTable
@@ -44,7 +44,8 @@ def run_once(gens, graph, fused):
     policy = Policy(nn, 0.02, Adam(len(Policy.get_flat(nn)), 0.01))
     nt = NoiseTable.create_shared(comm, 5_000_000, len(policy), seed=6, device=dev)
     rs = np.random.RandomState(88)
-    eng = GpuEngine(cfg, comm, policy, nt, env, rs, use_graph=graph, fused=fused)
+    eng = GpuEngine(cfg, comm, policy, nt, env, rs, use_graph=graph, fused=fused,
+                    pair_rollout=pair, eps_fp8=fp8)
     ranker = CenteredRanker()
     fits_hist = []
     for _ in range(gens):
@@ -63,13 +64,16 @@ def main():
     p.add_argument("--repeats", type=int, default=5)
     p.add_argument("--gens", type=int, default=3)
     p.add_argument("--no-graph", action="store_true")
+    p.add_argument("--pair", action="store_true")
+    p.add_argument("--fp8", action="store_true")
     args = p.parse_args()
     assert torch.cuda.is_available(), "race_check needs a GPU"
 
     base = None
     failures = 0
     for r in range(args.repeats):
-        out = run_once(args.gens, graph=not args.no_graph, fused=None)
+        out = run_once(args.gens, graph=not args.no_graph, fused=None,
+                       pair=args.pair or None, fp8=args.fp8 or None)
         if base is None:
             base = out
             continue
@@ -86,7 +90,8 @@ def main():
         print(f"RACE CHECK FAILED: {failures} mismatching repeats")
         return 1
     print(f"race check OK: {args.repeats} repeats bitwise-identical "
-          f"({args.gens} gens each, graph={not args.no_graph})")
+          f"({args.gens} gens each, graph={not args.no_graph}, "
+          f"pair={args.pair}, fp8={args.fp8})")
     return 0
 
 

@@ -78,8 +78,18 @@ def main():
     # obstat merge: all ranks agree after dist_inc
     counts = comm.allgather_obj(float(policy.obstat.count))
     assert len(set(counts)) == 1, counts
+
+    # -- phase 2: forced pair rollout + fp8 eps stream (the flagship path)
+    eng2 = GpuEngine(cfg, comm, policy, nt, env, rs, pair_rollout=True,
+                     eps_fp8=True, use_graph=False)
+    assert eng2.pair_rollout and eng2.eps_fp8, "fp8 pair path must engage"
+    for gen in range(2):
+        eng2.step(ranker)
+        hashes = comm.allgather_obj(
+            hashlib.sha256(eng2.theta.cpu().numpy().tobytes()).hexdigest())
+        assert len(set(hashes)) == 1, f"fp8 gen {gen}: rank params diverged"
     if comm.rank == 0:
-        print(f"TWIN-RANK-OK world={comm.size} gens=4 ppg={ppg}", flush=True)
+        print(f"TWIN-RANK-OK world={comm.size} gens=4+2fp8 ppg={ppg}", flush=True)
 
 
 if __name__ == "__main__":
