@@ -460,17 +460,32 @@ __device__ __forceinline__ void loco_dyn_partials_pair(
   __syncthreads();
 }
 
+// ES_DYN_EARLY (experiment knob): compute the A-matvec partials BEFORE the
+// policy forward, into a dedicated LDS slab — the A sweep needs only the
+// raw state, so its L2 latency then overlaps the forward's HBM weight
+// streaming deterministically within the block (instead of statistically
+// across blocks). Costs 16 KB LDS per block (one resident block at fp8's
+// 4-block occupancy); which wins is measured.
+#ifndef ES_DYN_EARLY
+#define ES_DYN_EARLY 0
+#endif
+
 // EB = uint16_t (bf16 eps rows) or uint8_t (fp8 row-pair-interleaved rows)
 template <typename EB>
 __device__ __forceinline__ void loco_pair_step_body(
     const MlpShape& sh, const LocoArgs& la, const LocoPtrs& P,
     const uint16_t* tb, const EB* eb, int bp, int bm, uint64_t salt,
     float* bufAp, float* bufAm, float* bufBp, float* bufBm, float* partial,
-    float* rawsP, float* rawsM, float* abufP, float* abufM) {
+    float* rawsP, float* rawsM, float* abufP, float* abufM,
+    float* partA = nullptr) {
   const int tid = threadIdx.x, nth = blockDim.x;
   loco_build_obs(la, P, bp, bufAp, rawsP, tid, nth);
   loco_build_obs(la, P, bm, bufAm, rawsM, tid, nth);
   __syncthreads();
+  const bool early = ES_DYN_EARLY && partA != nullptr && la.S % 8 == 0;
+  if (early)
+    loco_dyn_partials_pair(P.Am, la.S, rawsP, rawsM, partA, partA + 2048,
+                           tid, nth);
   float *ap, *am;
   if constexpr (sizeof(EB) == 1)
     mlp_layers_pair_fp8(tb, (const uint8_t*)eb, sh, bufAp, bufAm, bufBp, bufBm,
@@ -481,11 +496,12 @@ __device__ __forceinline__ void loco_pair_step_body(
   loco_decode_action(sh, la, P, bp, salt, ap, abufP, tid);
   loco_decode_action(sh, la, P, bm, salt, am, abufM, tid);
   __syncthreads();
-  if (la.S % 8 == 0)
+  float* pd = early ? partA : partial;
+  if (!early && la.S % 8 == 0)
     loco_dyn_partials_pair(P.Am, la.S, rawsP, rawsM, partial, partial + 2048,
                            tid, nth);
-  loco_dyn_finish(la, P, bp, rawsP, abufP, partial, tid, nth);
-  loco_dyn_finish(la, P, bm, rawsM, abufM, partial + 2048, tid, nth);
+  loco_dyn_finish(la, P, bp, rawsP, abufP, pd, tid, nth);
+  loco_dyn_finish(la, P, bm, rawsM, abufM, pd + 2048, tid, nth);
 }
 
 #define ES_LOCO_PAIR_CARVE()                                     \
@@ -499,6 +515,7 @@ __device__ __forceinline__ void loco_pair_step_body(
   float* rawsM = rawsP + ((la.S + 3) & ~3);                      \
   float* abufP = rawsM + ((la.S + 3) & ~3);                      \
   float* abufM = abufP + 64;                                     \
+  float* partA = ES_DYN_EARLY ? abufM + 64 : nullptr;            \
   const int q = blockIdx.x;                                      \
   const int p = q / la.eps, e = q % la.eps;                      \
   const int bp = p * la.eps + e;                                 \
@@ -515,7 +532,7 @@ loco_pair_step_kernel(MlpShape sh, LocoArgs la, LocoPtrs P, const uint16_t* tb,
                       const uint16_t* eb, int n_pairs, uint64_t salt) {
   ES_LOCO_PAIR_CARVE();
   loco_pair_step_body(sh, la, P, tb, ebp, bp, bm, salt, bufAp, bufAm, bufBp, bufBm,
-                      partial, rawsP, rawsM, abufP, abufM);
+                      partial, rawsP, rawsM, abufP, abufM, partA);
 }
 
 __global__ void __launch_bounds__(256, ES_PAIR_MINWAVES)
@@ -523,7 +540,7 @@ loco_pair_step_fp8_kernel(MlpShape sh, LocoArgs la, LocoPtrs P, const uint16_t* 
                           const uint8_t* eb, int n_pairs, uint64_t salt) {
   ES_LOCO_PAIR_CARVE();
   loco_pair_step_body(sh, la, P, tb, ebp, bp, bm, salt, bufAp, bufAm, bufBp, bufBm,
-                      partial, rawsP, rawsM, abufP, abufM);
+                      partial, rawsP, rawsM, abufP, abufM, partA);
 }
 
 // Whole-episode (or k-step chunk) pair rollout in ONE launch: each block
@@ -549,7 +566,7 @@ loco_pair_episode_kernel(MlpShape sh, LocoArgs la, LocoPtrs P, const uint16_t* t
   for (int t = 1; t <= n_steps; ++t)
     loco_pair_step_body(sh, la, P, tb, ebp, bp, bm, (uint64_t)(salt_base + t),
                         bufAp, bufAm, bufBp, bufBm, partial, rawsP, rawsM, abufP,
-                        abufM);
+                        abufM, partA);
 }
 
 // (A 128-thread 2-waves/SIMD variant — zero spills, 4 blocks/CU — was
@@ -694,7 +711,8 @@ extern "C" int es_loco_pair_step(
                          member_steps, behv, mo_sum, mo_sumsq);
   const int Spad = (sdim + 3) & ~3;
   const unsigned lds =
-      (unsigned)((4 * sh.maxdim + 2 * 256 * 8 + 2 * Spad + 2 * 64) * sizeof(float));
+      (unsigned)((4 * sh.maxdim + 2 * 256 * 8 + 2 * Spad + 2 * 64 +
+                  (ES_DYN_EARLY ? 2 * 2048 : 0)) * sizeof(float));
   const unsigned grid = (unsigned)(n_pairs * la.eps);
   loco_pair_step_kernel<<<dim3(grid), dim3(256), lds, (hipStream_t)stream>>>(
       sh, la, P, (const uint16_t*)theta_row, (const uint16_t*)eps_rows, n_pairs, salt);
@@ -727,7 +745,8 @@ extern "C" int es_loco_pair_step_fp8(
                          member_steps, behv, mo_sum, mo_sumsq);
   const int Spad = (sdim + 3) & ~3;
   const unsigned lds =
-      (unsigned)((4 * sh.maxdim + 2 * 256 * 8 + 2 * Spad + 2 * 64) * sizeof(float));
+      (unsigned)((4 * sh.maxdim + 2 * 256 * 8 + 2 * Spad + 2 * 64 +
+                  (ES_DYN_EARLY ? 2 * 2048 : 0)) * sizeof(float));
   const unsigned grid = (unsigned)(n_pairs * la.eps);
   loco_pair_step_fp8_kernel<<<dim3(grid), dim3(256), lds, (hipStream_t)stream>>>(
       sh, la, P, (const uint16_t*)theta_row, (const uint8_t*)eps_rows, n_pairs, salt);
@@ -761,7 +780,8 @@ extern "C" int es_loco_pair_episode(
                          member_steps, behv, mo_sum, mo_sumsq);
   const int Spad = (sdim + 3) & ~3;
   const unsigned lds =
-      (unsigned)((4 * sh.maxdim + 2 * 256 * 8 + 2 * Spad + 2 * 64) * sizeof(float));
+      (unsigned)((4 * sh.maxdim + 2 * 256 * 8 + 2 * Spad + 2 * 64 +
+                  (ES_DYN_EARLY ? 2 * 2048 : 0)) * sizeof(float));
   const unsigned grid = (unsigned)(n_pairs * la.eps);
   loco_pair_episode_kernel<<<dim3(grid), dim3(256), lds, (hipStream_t)stream>>>(
       sh, la, P, (const uint16_t*)theta_row, (const uint16_t*)eps_rows, n_pairs,
