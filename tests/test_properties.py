@@ -76,9 +76,11 @@ def test_obstat_merge_is_order_independent(parts):
 
     fwd = merged(range(len(parts)))
     rev = merged(reversed(range(len(parts))))
-    np.testing.assert_allclose(fwd.sum, rev.sum, rtol=1e-12, atol=1e-9)
-    np.testing.assert_allclose(fwd.sumsq, rev.sumsq, rtol=1e-12, atol=1e-9)
-    assert fwd.count == rev.count
+    # fp addition is order-sensitive at the last ulps; the merge contract is
+    # mathematical order-independence, asserted to ~1e-9 relative
+    np.testing.assert_allclose(fwd.sum, rev.sum, rtol=1e-9, atol=1e-6)
+    np.testing.assert_allclose(fwd.sumsq, rev.sumsq, rtol=1e-9, atol=1e-6)
+    np.testing.assert_allclose(fwd.count, rev.count, rtol=1e-12, atol=0)
 
 
 @given(st.lists(st.integers(1, 40), min_size=2, max_size=5))
