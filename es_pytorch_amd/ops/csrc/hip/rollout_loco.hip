@@ -413,6 +413,13 @@ static int loco_launch_dyn(const LocoArgs& la, const LocoPtrs& P, const float* a
 // verbatim, so per-slot bookkeeping is identical to the fused step.
 // Effective weights are bf16(theta) +- bf16(sigma*eps) (two roundings); with
 // sigma = 0 the trajectories are BITWISE-identical to es_loco_step.
+// ES_DYN_DEPTH8 (experiment knob): 8-deep lookahead on the shared-A L2
+// stream (12 uint4 live instead of 8; the fp8 step kernel has the VGPR
+// headroom). Accumulation order (ascending i) unchanged -> bitwise-same.
+#ifndef ES_DYN_DEPTH8
+#define ES_DYN_DEPTH8 0
+#endif
+
 __device__ __forceinline__ void loco_dyn_partials_pair(
     const uint16_t* Am, int S, const float* rawsP, const float* rawsM,
     float* partP, float* partM, int tid, int nth) {
@@ -433,6 +440,33 @@ __device__ __forceinline__ void loco_dyn_partials_pair(
     };
     int i = ip;
     const int step4 = PART * 4;
+#if ES_DYN_DEPTH8
+    if (i + 7 * PART < S) {
+      uint4 c0 = ld(i), c1 = ld(i + PART), c2 = ld(i + 2 * PART), c3 = ld(i + 3 * PART);
+      uint4 d0 = ld(i + 4 * PART), d1 = ld(i + 5 * PART), d2 = ld(i + 6 * PART),
+            d3 = ld(i + 7 * PART);
+      for (; i + 11 * PART < S; i += step4) {
+        const uint4 n0 = ld(i + 8 * PART), n1 = ld(i + 9 * PART),
+                    n2 = ld(i + 10 * PART), n3 = ld(i + 11 * PART);
+        fma2(c0, i);
+        fma2(c1, i + PART);
+        fma2(c2, i + 2 * PART);
+        fma2(c3, i + 3 * PART);
+        c0 = d0; c1 = d1; c2 = d2; c3 = d3;
+        d0 = n0; d1 = n1; d2 = n2; d3 = n3;
+      }
+      // drain the 8 preloaded blocks (guarded; order stays ascending)
+      if (i < S) fma2(c0, i);
+      if (i + PART < S) fma2(c1, i + PART);
+      if (i + 2 * PART < S) fma2(c2, i + 2 * PART);
+      if (i + 3 * PART < S) fma2(c3, i + 3 * PART);
+      if (i + 4 * PART < S) fma2(d0, i + 4 * PART);
+      if (i + 5 * PART < S) fma2(d1, i + 5 * PART);
+      if (i + 6 * PART < S) fma2(d2, i + 6 * PART);
+      if (i + 7 * PART < S) fma2(d3, i + 7 * PART);
+      i += 8 * PART;
+    }
+#else
     if (i + 3 * PART < S) {
       uint4 c0 = ld(i), c1 = ld(i + PART), c2 = ld(i + 2 * PART), c3 = ld(i + 3 * PART);
       for (; i + 7 * PART < S; i += step4) {
@@ -450,6 +484,7 @@ __device__ __forceinline__ void loco_dyn_partials_pair(
       fma2(c3, i + 3 * PART);
       i += step4;
     }
+#endif
     for (; i < S; i += PART) fma2(ld(i), i);
 #pragma unroll
     for (int q = 0; q < 8; ++q) {
