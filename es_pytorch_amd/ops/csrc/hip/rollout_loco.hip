@@ -19,6 +19,15 @@
 //    straggler block measured ~15-20% tail on every step).
 #include "mlp_core.h"
 
+// ES_DYN_DEPTH8: 8-deep lookahead on the shared-A L2 stream (12 uint4 live
+// instead of 8). Measured -4.3% same-box on the fp8 flagship (66.9 ->
+// 64.1-64.4 us/step); the fp8 kernel lands at 123 VGPRs (4 waves kept) and
+// the bf16 pair kernel stays 168 (its peak is in the forward). Default ON;
+// accumulation order (ascending i) unchanged -> bitwise-same results.
+#ifndef ES_DYN_DEPTH8
+#define ES_DYN_DEPTH8 1
+#endif
+
 struct LocoArgs {
   int S;               // latent state dim
   int A;               // action dim
@@ -129,6 +138,33 @@ __device__ __forceinline__ void loco_dyn_partials(
     };
     int i = ip;
     const int step4 = PART * 4;
+#if ES_DYN_DEPTH8
+    // 8-deep lookahead on the L2 A stream (see loco_dyn_partials_pair)
+    if (i + 7 * PART < S) {
+      uint4 c0 = ld(i), c1 = ld(i + PART), c2 = ld(i + 2 * PART), c3 = ld(i + 3 * PART);
+      uint4 d0 = ld(i + 4 * PART), d1 = ld(i + 5 * PART), d2 = ld(i + 6 * PART),
+            d3 = ld(i + 7 * PART);
+      for (; i + 11 * PART < S; i += step4) {
+        const uint4 n0 = ld(i + 8 * PART), n1 = ld(i + 9 * PART),
+                    n2 = ld(i + 10 * PART), n3 = ld(i + 11 * PART);
+        bf8_fma(c0, raws[i], acc);
+        bf8_fma(c1, raws[i + PART], acc);
+        bf8_fma(c2, raws[i + 2 * PART], acc);
+        bf8_fma(c3, raws[i + 3 * PART], acc);
+        c0 = d0; c1 = d1; c2 = d2; c3 = d3;
+        d0 = n0; d1 = n1; d2 = n2; d3 = n3;
+      }
+      if (i < S) bf8_fma(c0, raws[i], acc);
+      if (i + PART < S) bf8_fma(c1, raws[i + PART], acc);
+      if (i + 2 * PART < S) bf8_fma(c2, raws[i + 2 * PART], acc);
+      if (i + 3 * PART < S) bf8_fma(c3, raws[i + 3 * PART], acc);
+      if (i + 4 * PART < S) bf8_fma(d0, raws[i + 4 * PART], acc);
+      if (i + 5 * PART < S) bf8_fma(d1, raws[i + 5 * PART], acc);
+      if (i + 6 * PART < S) bf8_fma(d2, raws[i + 6 * PART], acc);
+      if (i + 7 * PART < S) bf8_fma(d3, raws[i + 7 * PART], acc);
+      i += 8 * PART;
+    }
+#else
     if (i + 3 * PART < S) {
       uint4 c0 = ld(i), c1 = ld(i + PART), c2 = ld(i + 2 * PART), c3 = ld(i + 3 * PART);
       for (; i + 7 * PART < S; i += step4) {
@@ -146,6 +182,7 @@ __device__ __forceinline__ void loco_dyn_partials(
       bf8_fma(c3, raws[i + 3 * PART], acc);
       i += step4;
     }
+#endif
     for (; i < S; i += PART) bf8_fma(ld(i), raws[i], acc);
 #pragma unroll
     for (int q = 0; q < 8; ++q) partial[(ip * OCT + oi) * 8 + q] = acc[q];
@@ -413,14 +450,6 @@ static int loco_launch_dyn(const LocoArgs& la, const LocoPtrs& P, const float* a
 // verbatim, so per-slot bookkeeping is identical to the fused step.
 // Effective weights are bf16(theta) +- bf16(sigma*eps) (two roundings); with
 // sigma = 0 the trajectories are BITWISE-identical to es_loco_step.
-// ES_DYN_DEPTH8: 8-deep lookahead on the shared-A L2 stream (12 uint4 live
-// instead of 8). Measured -4.3% same-box on the fp8 flagship (66.9 ->
-// 64.1-64.4 us/step); the fp8 kernel lands at 123 VGPRs (4 waves kept) and
-// the bf16 pair kernel stays 168 (its peak is in the forward). Default ON;
-// accumulation order (ascending i) unchanged -> bitwise-same results.
-#ifndef ES_DYN_DEPTH8
-#define ES_DYN_DEPTH8 1
-#endif
 
 __device__ __forceinline__ void loco_dyn_partials_pair(
     const uint16_t* Am, int S, const float* rawsP, const float* rawsM,
