@@ -46,7 +46,7 @@ def _interleave_map(dims, n):
     return out
 
 
-def _mk(dev, pop=64, steps=20, std=0.02, fp8=False, layers=(64, 64)):
+def _mk(dev, pop=64, steps=20, std=0.02, fp8=False, layers=(64, 64), eps=1):
     from es_pytorch_amd.config import AttrDict
     from es_pytorch_amd.core.engine import GpuEngine
     from es_pytorch_amd.core.noisetable import NoiseTable
@@ -64,8 +64,8 @@ def _mk(dev, pop=64, steps=20, std=0.02, fp8=False, layers=(64, 64)):
                                "l2coeff": 0.005, "lr": 0.01, "ob_clip": 5,
                                "save_obs_chance": 1.0},
                     "general": {"policies_per_gen": pop, "batch_size": 500,
-                                "seed": 3}})
-    env = make_batched("Humanoid-v2", pop + 1, dev, max_steps=steps,
+                                "seed": 3, "eps_per_policy": eps}})
+    env = make_batched("Humanoid-v2", (pop + 1) * eps, dev, max_steps=steps,
                        terminate_on_fall=False)
     nn = FeedForward(list(layers), torch.nn.Tanh(), env, 0.0, 5)
     policy = Policy(nn, std, Adam(len(Policy.get_flat(nn)), 0.01))
@@ -101,12 +101,14 @@ def test_pheno_fp8_layout_and_encoding(dev):
         assert not bad.any(), (p_i, np.argwhere(bad)[:5], want[bad][:5], got[bad][:5])
 
 
-def test_fp8_sigma0_matches_bf16_pair(dev):
+@pytest.mark.parametrize("eps", [1, 2])
+def test_fp8_sigma0_matches_bf16_pair(dev, eps):
     """At sigma=0 both eps formats encode exact zeros; trajectories agree up
-    to the fp8 path's different per-thread summation partition."""
+    to the fp8 path's different per-thread summation partition (also with
+    episode averaging, eps_per_policy=2)."""
     outs = {}
     for fp8 in (False, True):
-        eng = _mk(dev, std=0.0, fp8=fp8)
+        eng = _mk(dev, std=0.0, fp8=fp8, eps=eps)
         from es_pytorch_amd.utils.rankers import CenteredRanker
         eng.step(CenteredRanker())
         torch.cuda.synchronize(dev)
