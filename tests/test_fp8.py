@@ -143,6 +143,54 @@ def test_quantized_gather_matches_rollout_perturbation(dev):
                                rtol=1e-6, atol=1e-12)
 
 
+@pytest.mark.parametrize("kind", ["binned", "ig", "igm"])
+def test_fp8_act_modes_sigma0(dev, kind):
+    """fp8 with the K9 binned decode and both integrated-gaussian heads:
+    sigma=0 must agree with the bf16 pair path (the non-8-aligned heads run
+    the scalar fp8 layout — plain element-ordered bytes)."""
+    from es_pytorch_amd.config import AttrDict
+    from es_pytorch_amd.core.engine import GpuEngine
+    from es_pytorch_amd.core.noisetable import NoiseTable
+    from es_pytorch_amd.core.policy import Policy
+    from es_pytorch_amd.envs import make_batched
+    from es_pytorch_amd.nn.nn import (FFBinned, FFIntegGausAction,
+                                      FFIntegGausActionMulti)
+    from es_pytorch_amd.nn.optimizers import Adam
+    from es_pytorch_amd.parallel.comm import Comm
+    from es_pytorch_amd.utils.rankers import CenteredRanker
+
+    outs = {}
+    for fp8 in (False, True):
+        torch.manual_seed(31)
+        comm = Comm(dev)
+        cfg = AttrDict({"env": {"name": "Humanoid-v2", "max_steps": 15},
+                        "noise": {"tbl_size": 1_000_000, "std": 0.0},
+                        "policy": {"layer_sizes": [32], "ac_std": 0.0,
+                                   "l2coeff": 0.005, "lr": 0.01, "ob_clip": 5,
+                                   "save_obs_chance": 1.0},
+                        "general": {"policies_per_gen": 8, "batch_size": 500,
+                                    "seed": 6}})
+        env = make_batched("Humanoid-v2", 9, dev, max_steps=15,
+                           terminate_on_fall=False)
+        if kind == "binned":
+            nn = FFBinned([32], torch.nn.Tanh(), env, n_bins=5, ob_clip=5)
+        else:
+            cls = FFIntegGausAction if kind == "ig" else FFIntegGausActionMulti
+            nn = cls([32], torch.nn.Tanh(), env, ac_std=0.0, ob_clip=5)
+        policy = Policy(nn, 0.0, Adam(len(Policy.get_flat(nn)), 0.01))
+        nt = NoiseTable.create_shared(comm, 1_000_000, len(policy), seed=9,
+                                      device=dev)
+        eng = GpuEngine(cfg, comm, policy, nt, env, np.random.RandomState(7),
+                        use_graph=False, pair_rollout=True, eps_fp8=fp8)
+        if fp8:
+            assert eng.eps_fp8, kind
+        eng.step(CenteredRanker())
+        torch.cuda.synchronize(dev)
+        outs[fp8] = eng.rew_total.cpu().numpy().copy()
+    np.testing.assert_allclose(outs[True], outs[False], rtol=2e-4, atol=2e-3,
+                               err_msg=kind)
+
+
 def test_fp8_fidelity_at_flagship_sigma(dev):
     """sigma=0.02, 200 steps, pop 512: member fitness ranking and the
     reconstructed gradient must track the bf16 pair path closely."""
