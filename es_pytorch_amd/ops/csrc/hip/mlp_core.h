@@ -10,6 +10,10 @@
 #pragma once
 #include "common.h"
 
+#ifndef ES_FP8_WIDE4
+#define ES_FP8_WIDE4 0
+#endif
+
 #define ES_MAXL 8
 #define ES_MAXDIM 2048
 
@@ -265,6 +269,42 @@ __device__ __forceinline__ void mlp_layers_pair_fp8(
         };
         int i2 = ip;
         const int step2 = PART * 2;
+#if ES_FP8_WIDE4
+        // 4-wide (8 rows) pipeline: doubles the issue batch and lookahead
+        const int step4w = PART * 4;
+        if (i2 + 3 * PART < I2) {
+          uint4 e0 = lde8(i2), e1 = lde8(i2 + PART), e2 = lde8(i2 + 2 * PART),
+                e3 = lde8(i2 + 3 * PART);
+          uint4 t00 = ldt(2 * i2), t01 = ldt(2 * i2 + 1);
+          uint4 t10 = ldt(2 * (i2 + PART)), t11 = ldt(2 * (i2 + PART) + 1);
+          uint4 t20 = ldt(2 * (i2 + 2 * PART)), t21 = ldt(2 * (i2 + 2 * PART) + 1);
+          uint4 t30 = ldt(2 * (i2 + 3 * PART)), t31 = ldt(2 * (i2 + 3 * PART) + 1);
+          for (; i2 + 7 * PART < I2; i2 += step4w) {
+            const uint4 ne0 = lde8(i2 + 4 * PART), ne1 = lde8(i2 + 5 * PART),
+                        ne2 = lde8(i2 + 6 * PART), ne3 = lde8(i2 + 7 * PART);
+            const uint4 nt00 = ldt(2 * (i2 + 4 * PART)),
+                        nt01 = ldt(2 * (i2 + 4 * PART) + 1),
+                        nt10 = ldt(2 * (i2 + 5 * PART)),
+                        nt11 = ldt(2 * (i2 + 5 * PART) + 1),
+                        nt20 = ldt(2 * (i2 + 6 * PART)),
+                        nt21 = ldt(2 * (i2 + 6 * PART) + 1),
+                        nt30 = ldt(2 * (i2 + 7 * PART)),
+                        nt31 = ldt(2 * (i2 + 7 * PART) + 1);
+            fma_pairblk(e0, t00, t01, 2 * i2);
+            fma_pairblk(e1, t10, t11, 2 * (i2 + PART));
+            fma_pairblk(e2, t20, t21, 2 * (i2 + 2 * PART));
+            fma_pairblk(e3, t30, t31, 2 * (i2 + 3 * PART));
+            e0 = ne0; e1 = ne1; e2 = ne2; e3 = ne3;
+            t00 = nt00; t01 = nt01; t10 = nt10; t11 = nt11;
+            t20 = nt20; t21 = nt21; t30 = nt30; t31 = nt31;
+          }
+          fma_pairblk(e0, t00, t01, 2 * i2);
+          fma_pairblk(e1, t10, t11, 2 * (i2 + PART));
+          fma_pairblk(e2, t20, t21, 2 * (i2 + 2 * PART));
+          fma_pairblk(e3, t30, t31, 2 * (i2 + 3 * PART));
+          i2 += step4w;
+        }
+#else
         if (i2 + PART < I2) {
           // 2-wide (4 rows) software-pipelined double buffer, mirroring the
           // bf16 path's issue batching at the same register footprint
@@ -286,6 +326,7 @@ __device__ __forceinline__ void mlp_layers_pair_fp8(
           fma_pairblk(e1, t10, t11, 2 * (i2 + PART));
           i2 += step2;
         }
+#endif
         for (; i2 < I2; i2 += PART)
           fma_pairblk(lde8(i2), ldt(2 * i2), ldt(2 * i2 + 1), 2 * i2);
 #pragma unroll
