@@ -8,10 +8,12 @@ One rank = one GPU; per generation:
    index semantics, ``noisetable.py:37-40``) — one batched draw;
 2. HIP pheno kernel materializes the member parameters in HBM — either as
    per-member bf16 blobs [+noise pairs | -noise pairs | noiseless slot], or
-   (pair_rollout, the flagship default for pop >= 768) as ONE bf16 theta row
-   plus per-pair bf16 sigma*eps rows that the pair rollout kernel combines
-   in registers, halving the per-step HBM weight stream (the noiseless
-   evaluation of ``es.py:48`` rides in the same batch either way);
+   (pair_rollout, the flagship default) as ONE bf16 theta row plus per-pair
+   sigma*eps rows — fp8 e4m3 on supporting layouts (bytes AND load count
+   halve; the gradient gather re-quantizes through the same converters so
+   the update is estimator-exact), bf16 otherwise — that the pair rollout
+   kernel combines into W+/- in registers (the noiseless evaluation of
+   ``es.py:48`` rides in the same batch either way);
 3. the rollout loop — fused HIP MLP forward over the whole population +
    batched env dynamics — is captured once into a hipGraph and replayed
    every generation (launch-overhead-free inner loop; SURVEY.md §7.2 step 2);
