@@ -32,11 +32,17 @@ def make_batched(name: str, batch: int, device="cpu", **kwargs) -> BatchedEnv:
     if c in ("HumanoidFlagrun", "HumanoidFlagrunHarder"):
         kwargs.setdefault("goal_conditioned", True)
         return SyntheticLocomotion("HumanoidFlagrun", batch, device, **kwargs)
-    if c in ("AntFlagrun", "AntGather"):  # reference flagrun.py's AntGather env
+    if c in ("AntFlagrun", "AntGather", "AntMaze"):
+        # reference flagrun.py's AntGather and ns.json's AntMaze (both
+        # hrl_pybullet_envs): goal/target-directed Ant stand-ins
         kwargs.setdefault("goal_conditioned", True)
         return SyntheticLocomotion("AntFlagrun", batch, device, **kwargs)
     if c in LOCO_SHAPES:
         return SyntheticLocomotion(c, batch, device, **kwargs)
+    # case-insensitive fallback: PyBullet spells it Walker2D, MuJoCo Walker2d
+    for k in LOCO_SHAPES:
+        if k.lower() == c.lower():
+            return SyntheticLocomotion(k, batch, device, **kwargs)
     raise ValueError(f"unknown env: {name!r} (canonical {c!r}); "
                      f"known: CartPole, Pendulum, {sorted(LOCO_SHAPES)}")
 
