@@ -78,3 +78,33 @@ def test_reference_config_env_names_resolve():
     for n in sorted(names):
         env = make_batched(n, 2, torch.device("cpu"), max_steps=5)
         assert env.ob_dim > 0 and env.ac_dim > 0, n
+
+
+def test_reference_configs_assemble():
+    """The reference's OWN config files must drive this framework: load via
+    our config loader, resolve the env, size the network, and build the
+    Policy — the switch-over contract beyond just env names."""
+    import glob
+    import json
+
+    from es_pytorch_amd.config import load_config
+    from es_pytorch_amd.core.policy import Policy
+    from es_pytorch_amd.envs import make
+    from es_pytorch_amd.nn.nn import FeedForward
+    from es_pytorch_amd.nn.optimizers import Adam
+
+    paths = sorted(glob.glob("/root/reference/configs/*.json"))
+    if not paths:
+        pytest.skip("reference repo not present on this machine")
+    for p in paths:
+        cfg = load_config(p)
+        if "env" not in cfg:  # batch.json is a sweep spec, not an experiment
+            continue
+        env = make(cfg.env.name, max_steps=min(10, int(cfg.env.max_steps)))
+        nn = FeedForward(cfg.policy.layer_sizes, torch.nn.Tanh(), env,
+                         cfg.policy.ac_std, cfg.policy.ob_clip)
+        policy = Policy(nn, cfg.noise.std, Adam(len(Policy.get_flat(nn)),
+                                                cfg.policy.lr))
+        assert len(policy) > 0, p
+        # the generation knobs our entry scripts read must be present
+        assert cfg.general.policies_per_gen > 0 and cfg.general.gens > 0, p
