@@ -100,7 +100,8 @@ def test_reference_configs_assemble():
         cfg = load_config(p)
         if "env" not in cfg:  # batch.json is a sweep spec, not an experiment
             continue
-        env = make(cfg.env.name, max_steps=min(10, int(cfg.env.max_steps)))
+        env = make(cfg.env.name,
+                   max_steps=min(10, int(cfg.env.get("max_steps", 1000))))
         nn = FeedForward(cfg.policy.layer_sizes, torch.nn.Tanh(), env,
                          cfg.policy.ac_std, cfg.policy.ob_clip)
         policy = Policy(nn, cfg.noise.std, Adam(len(Policy.get_flat(nn)),
