@@ -406,7 +406,8 @@ class GpuEngine:
 
     def _loco_episode(self, member_base: int, n_members: int, noiseless_from: int,
                       n_steps: Optional[int] = None, salt_base: int = 0,
-                      weights_ptr: Optional[int] = None, wrow0: int = 0):
+                      weights_ptr: Optional[int] = None, wrow0: int = 0,
+                      block_threads: int = 256):
         """n_steps consecutive env steps (default: the whole episode) for
         [member_base, member_base+n_members) slots in one launch. weights_ptr
         + wrow0 let the call run on a sub-blob (pair mode: the 1-row theta
@@ -430,8 +431,8 @@ class GpuEngine:
             int(env.goal_conditioned), int(env.terminate_on_fall), noiseless_from,
             self.bins, self.eps, self.act_mode,
             float(env.leak), float(env.ctrl_cost), float(env.alive_bonus),
-            float(env.fall_threshold), float(env.dt), self._stream()),
-            "es_loco_episode")
+            float(env.fall_threshold), float(env.dt), int(block_threads),
+            self._stream()), "es_loco_episode")
 
     def _loco_pair_episode(self, n_steps: Optional[int] = None, salt_base: int = 0):
         """n_steps consecutive env steps for every (pair, episode) block in
@@ -464,12 +465,16 @@ class GpuEngine:
             self._stream()), "es_loco_pair_episode")
 
     def _loco_noiseless_episode(self):
+        # 512-thread blocks: the single-member episode is latency-serial, so
+        # doubling PART halves its dependency chains (the population grids
+        # keep 256-thread blocks for occupancy)
         if self.pair_rollout:
             self._loco_episode((self.M - 1) * self.eps, self.eps, 0,
                                weights_ptr=self.theta_row.data_ptr(),
-                               wrow0=self.M - 1)
+                               wrow0=self.M - 1, block_threads=512)
         else:
-            self._loco_episode((self.M - 1) * self.eps, self.eps, 0)
+            self._loco_episode((self.M - 1) * self.eps, self.eps, 0,
+                               block_threads=512)
 
     # ------------------------------------------------------------- rollout
     def _step_body(self, t: int):

@@ -198,7 +198,7 @@ def _bind_hip(lib):
                                     p, p, p, p, p, p, p, p, p, p,
                                     p, p, p, p, p, p,
                                     i32, i32, i32, i32, i32, i32, i32, i32, i32, i32,
-                                    i32, i32, f32, f32, f32, f32, f32, p]
+                                    i32, i32, f32, f32, f32, f32, f32, i32, p]
     for fn in ["es_noise_fill", "es_pheno_bf16", "es_pheno_fp8", "es_mlp_fwd",
                "es_grad_gather", "es_adam_step", "es_sgd_step", "es_loco_step",
                "es_loco_step_split", "es_loco_pair_step", "es_loco_pair_step_fp8",

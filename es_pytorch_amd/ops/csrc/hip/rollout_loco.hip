@@ -332,6 +332,27 @@ loco_episode_kernel(MlpShape sh, LocoArgs la, LocoPtrs P, int member_base, int n
                    abuf, sc);
 }
 
+// 512-thread variant for the LATENCY-BOUND noiseless side episode: one
+// block runs max_steps sequential steps, so halving every per-step
+// partition (PART doubles) shortens the serial dependency chains. Used
+// only for the noiseless member (block_threads=512) — the episode-mode
+// population grid keeps 256-thread blocks for occupancy.
+__global__ void __launch_bounds__(512)
+loco_episode512_kernel(MlpShape sh, LocoArgs la, LocoPtrs P, int member_base,
+                       int n_steps, int salt_base) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* bufA = reinterpret_cast<float*>(smem);
+  float* bufB = bufA + sh.maxdim;
+  float* partial = bufB + sh.maxdim;
+  float* raws = partial + 512 * 8;
+  float* abuf = raws + ((la.S + 3) & ~3);
+  float* sc = abuf + 64;
+  const int b = member_base + blockIdx.x;
+  for (int t = 1; t <= n_steps; ++t)
+    loco_step_body(sh, la, P, b, (uint64_t)(salt_base + t), bufA, bufB, partial, raws,
+                   abuf, sc);
+}
+
 // ---- split-dynamics path ---------------------------------------------------
 // The fused kernel re-reads the shared (S, S) bf16 transition matrix A once
 // per MEMBER (pop x 283 KB per step for Humanoid = the measured ~20 us/step
@@ -870,7 +891,7 @@ extern "C" int es_loco_episode(const void* weights, const void* obmean, const vo
                                int32_t goal_flag, int32_t terminate, int32_t noiseless_from,
                                int32_t bins, int32_t eps, int32_t act_mode, float leak,
                                float ctrl, float alive_bonus, float fall_thr, float dt,
-                               void* stream) {
+                               int32_t block_threads, void* stream) {
   MlpShape sh;
   LocoArgs la;
   unsigned lds;
@@ -882,8 +903,16 @@ extern "C" int es_loco_episode(const void* weights, const void* obmean, const vo
   LocoPtrs P = loco_ptrs(weights, obmean, obstd, ac_std_dev, seed_dev, s_glob, pos, goal,
                          Am, Bm, b0, wv, wa, wy, wh, alive, rew_total, member_steps, behv,
                          mo_sum, mo_sumsq);
-  loco_episode_kernel<<<dim3((unsigned)n_members), dim3(256), lds, (hipStream_t)stream>>>(
-      sh, la, P, member_base, n_steps, salt_base);
+  if (block_threads == 512) {
+    const unsigned lds512 = lds + 256 * 8 * 4;  // partial[] grows with nth
+    loco_episode512_kernel<<<dim3((unsigned)n_members), dim3(512), lds512,
+                             (hipStream_t)stream>>>(sh, la, P, member_base, n_steps,
+                                                    salt_base);
+  } else {
+    loco_episode_kernel<<<dim3((unsigned)n_members), dim3(256), lds,
+                          (hipStream_t)stream>>>(sh, la, P, member_base, n_steps,
+                                                 salt_base);
+  }
   ES_CHECK_LAUNCH();
   return 0;
 }
