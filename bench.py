@@ -13,8 +13,13 @@ Launch (driver contract):
 
 A "step" is one ES generation. Data is synthetic (seeded latent-dynamics env
 of Humanoid-v2's 376-obs/17-action shape; no physics engine or datasets exist
-offline) with random-init weights; compute dtype bf16 (weights/forward) with
-fp32 master params and fp64 fitness exchange.
+offline) with random-init weights; compute dtype bf16 (theta/forward, fp32
+accumulation, fp32 master params, fp64 fitness exchange). The perturbation
+sigma*eps rows are encoded e4m3 by default (reported as config.eps_encoding;
+--no-fp8 reverts to bf16): the gradient gather re-quantizes through the same
+hardware converters, so the update is estimator-exact ES on the quantized
+perturbation distribution — fidelity, learning and TCC evidence in
+profiles/README.md.
 """
 from __future__ import annotations
 
