@@ -15,7 +15,7 @@ reporter is covered without mlflow.
 """
 from __future__ import annotations
 
-from typing import List, Optional, Tuple
+from typing import List, Optional
 
 import numpy as np
 

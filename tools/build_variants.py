@@ -15,7 +15,6 @@ sources (checked out into a temp tree):
 """
 import argparse
 import os
-import shutil
 import subprocess
 import sys
 import tempfile
