@@ -4,12 +4,14 @@ The reference never tested these (SURVEY.md §4 gap); asserted here against
 independently hand-computed updates.
 """
 import numpy as np
+import pytest
 
 from es_pytorch_amd.nn.optimizers import SGD, Adam, SimpleES
 
 
 def test_simple_es():
-    o = SimpleES(3, lr=0.5)
+    with pytest.warns(UserWarning, match="sign convention|descends"):
+        o = SimpleES(3, lr=0.5)
     g = np.array([1.0, -2.0, 3.0], dtype=np.float32)
     np.testing.assert_allclose(o.step(g), 0.5 * g)
     assert o.t == 1

@@ -70,7 +70,8 @@ def test_reference_config_env_names_resolve():
     names = set()
     for p in glob.glob("/root/reference/configs/*.json"):
         try:
-            names.add(json.load(open(p))["env"]["name"])
+            with open(p) as f:
+                names.add(json.load(f)["env"]["name"])
         except Exception:
             pass
     if not names:
