@@ -287,12 +287,13 @@ class GpuEngine:
     @staticmethod
     def _fp8_layout_ok(dims: List[int]) -> bool:
         """Mirror of the fp8 row-pair interleave constraints (pheno.hip):
-        every vectorizable layer needs an even input dim and a 16-byte-
-        aligned weight offset so the 16-fp8 loads stay aligned."""
+        each vectorizable layer's PAIR REGION (rows after the plain first
+        row of odd input dims) must start 16-byte-aligned so the 16-fp8
+        loads stay aligned."""
         off = 0
         for I, O in zip(dims[:-1], dims[1:]):
             vec = (O % 8 == 0) and (off % 8 == 0)
-            if vec and (I % 2 != 0 or off % 16 != 0):
+            if vec and (off + (I % 2) * O) % 16 != 0:
                 return False
             off += I * O + O
         return True

@@ -234,7 +234,7 @@ __device__ __forceinline__ void mlp_layers_pair_fp8(
     const uint8_t* EBs8 = eb8 + sh.boff[l];
     const bool do_act = (l < sh.n_layers - 1) || act_final;
 
-    if (sh.vec_ok[l] && !(I & 1)) {
+    if (sh.vec_ok[l]) {
       const int OCT = O >> 3;
       const int PART = nthreads / OCT;
       const int oi = tid % OCT, ip = tid / OCT;
@@ -242,10 +242,11 @@ __device__ __forceinline__ void mlp_layers_pair_fp8(
 #pragma unroll
       for (int q = 0; q < 8; ++q) accp[q] = accm[q] = 0.0f;
       if (ip < PART) {
-        const int I2 = I >> 1;           // row pairs
+        const int ro = I & 1;            // odd input dim: row 0 stays plain
+        const int I2 = (I - ro) >> 1;    // row pairs (rows ro..I-1)
         const int rs2 = 2 * O;           // bytes per row pair
         const uint16_t* tcol = Tt + (oi << 3);
-        const uint8_t* ecol8 = Et8 + (oi << 4);
+        const uint8_t* ecol8 = Et8 + (int64_t)ro * O + (oi << 4);
         typedef uint32_t u32x4v __attribute__((ext_vector_type(4)));
         auto lde8 = [&](int i2) {  // 16 fp8 = octet of rows 2*i2, 2*i2+1
           u32x4v v = __builtin_nontemporal_load(
@@ -267,6 +268,15 @@ __device__ __forceinline__ void mlp_layers_pair_fp8(
           fp8x4_decode(e16.w, e8 + 4);
           bf8e_fma_pair(tb_, e8, xp[r0 + 1], xm[r0 + 1], accp, accm);
         };
+        if (ro && ip == 0) {
+          // row 0 (plain element-ordered fp8): every oi-octet lane of the
+          // ip==0 partition consumes it so the partial layout is unchanged
+          float e8[8];
+          const uint8_t* p0 = Et8 + (oi << 3);
+          fp8x4_decode(*reinterpret_cast<const uint32_t*>(p0), e8);
+          fp8x4_decode(*reinterpret_cast<const uint32_t*>(p0 + 4), e8 + 4);
+          bf8e_fma_pair(ldt(0), e8, xp[0], xm[0], accp, accm);
+        }
         int i2 = ip;
         const int step2 = PART * 2;
 #if ES_FP8_WIDE4
@@ -275,33 +285,33 @@ __device__ __forceinline__ void mlp_layers_pair_fp8(
         if (i2 + 3 * PART < I2) {
           uint4 e0 = lde8(i2), e1 = lde8(i2 + PART), e2 = lde8(i2 + 2 * PART),
                 e3 = lde8(i2 + 3 * PART);
-          uint4 t00 = ldt(2 * i2), t01 = ldt(2 * i2 + 1);
-          uint4 t10 = ldt(2 * (i2 + PART)), t11 = ldt(2 * (i2 + PART) + 1);
-          uint4 t20 = ldt(2 * (i2 + 2 * PART)), t21 = ldt(2 * (i2 + 2 * PART) + 1);
-          uint4 t30 = ldt(2 * (i2 + 3 * PART)), t31 = ldt(2 * (i2 + 3 * PART) + 1);
+          uint4 t00 = ldt(ro + 2 * i2), t01 = ldt(ro + 2 * i2 + 1);
+          uint4 t10 = ldt(ro + 2 * (i2 + PART)), t11 = ldt(ro + 2 * (i2 + PART) + 1);
+          uint4 t20 = ldt(ro + 2 * (i2 + 2 * PART)), t21 = ldt(ro + 2 * (i2 + 2 * PART) + 1);
+          uint4 t30 = ldt(ro + 2 * (i2 + 3 * PART)), t31 = ldt(ro + 2 * (i2 + 3 * PART) + 1);
           for (; i2 + 7 * PART < I2; i2 += step4w) {
             const uint4 ne0 = lde8(i2 + 4 * PART), ne1 = lde8(i2 + 5 * PART),
                         ne2 = lde8(i2 + 6 * PART), ne3 = lde8(i2 + 7 * PART);
-            const uint4 nt00 = ldt(2 * (i2 + 4 * PART)),
-                        nt01 = ldt(2 * (i2 + 4 * PART) + 1),
-                        nt10 = ldt(2 * (i2 + 5 * PART)),
-                        nt11 = ldt(2 * (i2 + 5 * PART) + 1),
-                        nt20 = ldt(2 * (i2 + 6 * PART)),
-                        nt21 = ldt(2 * (i2 + 6 * PART) + 1),
-                        nt30 = ldt(2 * (i2 + 7 * PART)),
-                        nt31 = ldt(2 * (i2 + 7 * PART) + 1);
-            fma_pairblk(e0, t00, t01, 2 * i2);
-            fma_pairblk(e1, t10, t11, 2 * (i2 + PART));
-            fma_pairblk(e2, t20, t21, 2 * (i2 + 2 * PART));
-            fma_pairblk(e3, t30, t31, 2 * (i2 + 3 * PART));
+            const uint4 nt00 = ldt(ro + 2 * (i2 + 4 * PART)),
+                        nt01 = ldt(ro + 2 * (i2 + 4 * PART) + 1),
+                        nt10 = ldt(ro + 2 * (i2 + 5 * PART)),
+                        nt11 = ldt(ro + 2 * (i2 + 5 * PART) + 1),
+                        nt20 = ldt(ro + 2 * (i2 + 6 * PART)),
+                        nt21 = ldt(ro + 2 * (i2 + 6 * PART) + 1),
+                        nt30 = ldt(ro + 2 * (i2 + 7 * PART)),
+                        nt31 = ldt(ro + 2 * (i2 + 7 * PART) + 1);
+            fma_pairblk(e0, t00, t01, ro + 2 * i2);
+            fma_pairblk(e1, t10, t11, ro + 2 * (i2 + PART));
+            fma_pairblk(e2, t20, t21, ro + 2 * (i2 + 2 * PART));
+            fma_pairblk(e3, t30, t31, ro + 2 * (i2 + 3 * PART));
             e0 = ne0; e1 = ne1; e2 = ne2; e3 = ne3;
             t00 = nt00; t01 = nt01; t10 = nt10; t11 = nt11;
             t20 = nt20; t21 = nt21; t30 = nt30; t31 = nt31;
           }
-          fma_pairblk(e0, t00, t01, 2 * i2);
-          fma_pairblk(e1, t10, t11, 2 * (i2 + PART));
-          fma_pairblk(e2, t20, t21, 2 * (i2 + 2 * PART));
-          fma_pairblk(e3, t30, t31, 2 * (i2 + 3 * PART));
+          fma_pairblk(e0, t00, t01, ro + 2 * i2);
+          fma_pairblk(e1, t10, t11, ro + 2 * (i2 + PART));
+          fma_pairblk(e2, t20, t21, ro + 2 * (i2 + 2 * PART));
+          fma_pairblk(e3, t30, t31, ro + 2 * (i2 + 3 * PART));
           i2 += step4w;
         }
 #else
@@ -309,26 +319,26 @@ __device__ __forceinline__ void mlp_layers_pair_fp8(
           // 2-wide (4 rows) software-pipelined double buffer, mirroring the
           // bf16 path's issue batching at the same register footprint
           uint4 e0 = lde8(i2), e1 = lde8(i2 + PART);
-          uint4 t00 = ldt(2 * i2), t01 = ldt(2 * i2 + 1);
-          uint4 t10 = ldt(2 * (i2 + PART)), t11 = ldt(2 * (i2 + PART) + 1);
+          uint4 t00 = ldt(ro + 2 * i2), t01 = ldt(ro + 2 * i2 + 1);
+          uint4 t10 = ldt(ro + 2 * (i2 + PART)), t11 = ldt(ro + 2 * (i2 + PART) + 1);
           for (; i2 + 3 * PART < I2; i2 += step2) {
             const uint4 ne0 = lde8(i2 + 2 * PART), ne1 = lde8(i2 + 3 * PART);
-            const uint4 nt00 = ldt(2 * (i2 + 2 * PART)),
-                        nt01 = ldt(2 * (i2 + 2 * PART) + 1),
-                        nt10 = ldt(2 * (i2 + 3 * PART)),
-                        nt11 = ldt(2 * (i2 + 3 * PART) + 1);
-            fma_pairblk(e0, t00, t01, 2 * i2);
-            fma_pairblk(e1, t10, t11, 2 * (i2 + PART));
+            const uint4 nt00 = ldt(ro + 2 * (i2 + 2 * PART)),
+                        nt01 = ldt(ro + 2 * (i2 + 2 * PART) + 1),
+                        nt10 = ldt(ro + 2 * (i2 + 3 * PART)),
+                        nt11 = ldt(ro + 2 * (i2 + 3 * PART) + 1);
+            fma_pairblk(e0, t00, t01, ro + 2 * i2);
+            fma_pairblk(e1, t10, t11, ro + 2 * (i2 + PART));
             e0 = ne0; e1 = ne1;
             t00 = nt00; t01 = nt01; t10 = nt10; t11 = nt11;
           }
-          fma_pairblk(e0, t00, t01, 2 * i2);
-          fma_pairblk(e1, t10, t11, 2 * (i2 + PART));
+          fma_pairblk(e0, t00, t01, ro + 2 * i2);
+          fma_pairblk(e1, t10, t11, ro + 2 * (i2 + PART));
           i2 += step2;
         }
 #endif
         for (; i2 < I2; i2 += PART)
-          fma_pairblk(lde8(i2), ldt(2 * i2), ldt(2 * i2 + 1), 2 * i2);
+          fma_pairblk(lde8(i2), ldt(ro + 2 * i2), ldt(ro + 2 * i2 + 1), ro + 2 * i2);
 #pragma unroll
         for (int q = 0; q < 8; ++q) {
           partial[(ip * OCT + oi) * 8 + q] = accp[q];

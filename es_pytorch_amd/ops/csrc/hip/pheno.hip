@@ -59,14 +59,17 @@ __global__ void pheno_bf16_kernel(uint16_t* __restrict__ out, const float* __res
 __device__ __forceinline__ int64_t fp8_src_elem(const MlpShape& sh, int64_t b) {
   for (int l = 0; l < sh.n_layers; ++l) {
     if (b < sh.boff[l]) {  // weight block l (b >= woff[l] by construction)
-      if (!sh.vec_ok[l] || (sh.dims[l] & 1)) return b;  // plain layout
+      if (!sh.vec_ok[l]) return b;  // scalar-path layer: plain layout
       const int O = sh.dims[l + 1];
-      const int64_t rel = b - sh.woff[l];
+      // odd input dim: row 0 stays plain; rows 1.. are pair-interleaved
+      const int ro = sh.dims[l] & 1;
+      const int64_t rel = b - sh.woff[l] - (int64_t)ro * O;
+      if (rel < 0) return b;  // inside the plain row 0
       const int64_t pairblk = rel / (2 * O);
       const int within = (int)(rel % (2 * O));
       const int col = (within >> 4) * 8 + (within & 7);
       const int half = (within >> 3) & 1;
-      return sh.woff[l] + (2 * pairblk + half) * (int64_t)O + col;
+      return sh.woff[l] + (ro + 2 * pairblk + half) * (int64_t)O + col;
     }
     const int64_t bias_end = sh.boff[l] + sh.dims[l + 1];
     if (b < bias_end) return b;  // bias: plain

@@ -37,16 +37,18 @@ def _interleave_map(dims, n):
     off = 0
     for I, O in zip(dims[:-1], dims[1:]):
         vec = (O % 8 == 0) and (off % 8 == 0)
-        if vec and I % 2 == 0:
-            e = np.arange(I * O, dtype=np.int64)
-            i, o = e // O, e % O
-            byte = (i & ~1) * O + (o >> 3) * 16 + (i & 1) * 8 + (o & 7)
-            out[off:off + I * O] = off + byte
+        if vec:
+            ro = I % 2  # odd input dim: row 0 stays plain
+            e = np.arange(ro * O, I * O, dtype=np.int64)
+            i, o = e // O - ro, e % O
+            byte = ro * O + (i & ~1) * O + (o >> 3) * 16 + (i & 1) * 8 + (o & 7)
+            out[off + ro * O: off + I * O] = off + byte
         off += I * O + O  # bias block stays identity
     return out
 
 
-def _mk(dev, pop=64, steps=20, std=0.02, fp8=False, layers=(64, 64), eps=1):
+def _mk(dev, pop=64, steps=20, std=0.02, fp8=False, layers=(64, 64), eps=1,
+        env_name="Humanoid-v2"):
     from es_pytorch_amd.config import AttrDict
     from es_pytorch_amd.core.engine import GpuEngine
     from es_pytorch_amd.core.noisetable import NoiseTable
@@ -58,14 +60,14 @@ def _mk(dev, pop=64, steps=20, std=0.02, fp8=False, layers=(64, 64), eps=1):
 
     torch.manual_seed(21)
     comm = Comm(dev)
-    cfg = AttrDict({"env": {"name": "Humanoid-v2", "max_steps": steps},
+    cfg = AttrDict({"env": {"name": env_name, "max_steps": steps},
                     "noise": {"tbl_size": 2_000_000, "std": std},
                     "policy": {"layer_sizes": list(layers), "ac_std": 0.0,
                                "l2coeff": 0.005, "lr": 0.01, "ob_clip": 5,
                                "save_obs_chance": 1.0},
                     "general": {"policies_per_gen": pop, "batch_size": 500,
                                 "seed": 3, "eps_per_policy": eps}})
-    env = make_batched("Humanoid-v2", (pop + 1) * eps, dev, max_steps=steps,
+    env = make_batched(env_name, (pop + 1) * eps, dev, max_steps=steps,
                        terminate_on_fall=False)
     nn = FeedForward(list(layers), torch.nn.Tanh(), env, 0.0, 5)
     policy = Policy(nn, std, Adam(len(Policy.get_flat(nn)), 0.01))
@@ -76,11 +78,16 @@ def _mk(dev, pop=64, steps=20, std=0.02, fp8=False, layers=(64, 64), eps=1):
     return eng
 
 
-def test_pheno_fp8_layout_and_encoding(dev):
+@pytest.mark.parametrize("env_name,layers", [
+    ("Humanoid-v2", (64, 64)),
+    # odd input dim (AntFlagrun obs = 29): row 0 plain, rows 1.. paired
+    ("AntFlagrunBulletEnv-v0", (64, 64)),
+])
+def test_pheno_fp8_layout_and_encoding(dev, env_name, layers):
     """The fp8 blob, de-interleaved and decoded on the host, must equal
     e4m3fn(sigma * table[offset + t]) elementwise."""
-    eng = _mk(dev, fp8=True)
-    assert eng.eps_fp8, "fp8 gate must accept the Humanoid MLP layout"
+    eng = _mk(dev, fp8=True, layers=layers, env_name=env_name)
+    assert eng.eps_fp8, f"fp8 gate must accept {env_name}"
     eng._upload_offsets()
     eng._pheno()
     torch.cuda.synchronize(dev)
@@ -101,14 +108,17 @@ def test_pheno_fp8_layout_and_encoding(dev):
         assert not bad.any(), (p_i, np.argwhere(bad)[:5], want[bad][:5], got[bad][:5])
 
 
-@pytest.mark.parametrize("eps", [1, 2])
-def test_fp8_sigma0_matches_bf16_pair(dev, eps):
+@pytest.mark.parametrize("eps,env_name", [(1, "Humanoid-v2"), (2, "Humanoid-v2"),
+                                          (1, "AntFlagrunBulletEnv-v0")])
+def test_fp8_sigma0_matches_bf16_pair(dev, eps, env_name):
     """At sigma=0 both eps formats encode exact zeros; trajectories agree up
     to the fp8 path's different per-thread summation partition (also with
     episode averaging, eps_per_policy=2)."""
     outs = {}
     for fp8 in (False, True):
-        eng = _mk(dev, std=0.0, fp8=fp8, eps=eps)
+        eng = _mk(dev, std=0.0, fp8=fp8, eps=eps, env_name=env_name)
+        if env_name != "Humanoid-v2":
+            assert eng.eps_fp8 == fp8
         from es_pytorch_amd.utils.rankers import CenteredRanker
         eng.step(CenteredRanker())
         torch.cuda.synchronize(dev)
