@@ -110,3 +110,22 @@ def test_noise_table_slices_match_full_fill(seed, k):
     idx = (seed % 7919) % (10_000 - 100 * k)
     np.testing.assert_array_equal(a.get(idx, 100 * k).numpy(),
                                   b.get(idx, 100 * k).numpy())
+
+
+@given(st.lists(st.integers(1, 33), min_size=2, max_size=4))
+@settings(max_examples=60, deadline=None)
+def test_fp8_interleave_map_is_a_bijection(dims):
+    """The e4m3 blob layout (pheno.hip fp8_src_elem, mirrored by the test
+    helper) must be a PERMUTATION of [0, n) for every layer shape — even,
+    odd and scalar-path layers alike; a collision or gap would silently
+    corrupt perturbations."""
+    import importlib.util
+    import os
+    spec = importlib.util.spec_from_file_location(
+        "fp8_test_helpers",
+        os.path.join(os.path.dirname(__file__), "test_fp8.py"))
+    mod = importlib.util.module_from_spec(spec)
+    spec.loader.exec_module(mod)
+    n = sum(i * o + o for i, o in zip(dims[:-1], dims[1:]))
+    m = mod._interleave_map(dims, n)
+    assert np.array_equal(np.sort(m), np.arange(n)), dims
