@@ -7,7 +7,7 @@ communicator setup and kernel launches of the exact transport the driver's
 multi-GPU run uses. Failures here (missing librccl, HSA IPC config, comm
 init hangs) would otherwise only surface mid-scale-run."""
 import datetime
-import os
+import socket
 
 import numpy as np
 import pytest
@@ -21,7 +21,10 @@ pytestmark = pytest.mark.gpu
 def test_rccl_initializes_and_reduces():
     if dist.is_initialized():
         pytest.skip("a process group already exists in this process")
-    store = dist.TCPStore("127.0.0.1", 29997, 1, True,
+    with socket.socket() as sock:  # dynamically allocated free port
+        sock.bind(("127.0.0.1", 0))
+        port = sock.getsockname()[1]
+    store = dist.TCPStore("127.0.0.1", port, 1, True,
                           datetime.timedelta(seconds=60))
     dist.init_process_group("nccl", store=store, rank=0, world_size=1,
                             timeout=datetime.timedelta(seconds=60))

@@ -9,6 +9,7 @@ kernels, rollouts, update) runs on the GPU — exactly the code path of an
 """
 import json
 import os
+import socket
 import subprocess
 import sys
 
@@ -20,13 +21,19 @@ pytestmark = pytest.mark.gpu
 REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
 
+def _free_port() -> int:
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        return s.getsockname()[1]
+
+
 def _torchrun(args, nproc=2, timeout=420):
     env = dict(os.environ)
     env.update({"ES_COMM_BACKEND": "gloo", "MASTER_ADDR": "127.0.0.1",
                 "HSA_ENABLE_IPC_MODE_LEGACY": env.get("HSA_ENABLE_IPC_MODE_LEGACY", "0")})
     cmd = [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
            f"--nproc-per-node={nproc}", "--master-addr", "127.0.0.1",
-           "--master-port", "29771", *args]
+           "--master-port", str(_free_port()), *args]
     return subprocess.run(cmd, cwd=REPO, env=env, capture_output=True, text=True,
                           timeout=timeout)
 
