@@ -125,3 +125,27 @@ def _broadcast_tensor_worker(rank, world):
 def test_broadcast_tensor_mp():
     """Comm.broadcast_tensor_ (the archive-growth primitive) across 2 ranks."""
     assert all(run_mp(_broadcast_tensor_worker, world=2))
+
+
+def test_comm_single_process_fallbacks():
+    """With no initialized process group, every collective must degrade to
+    a correct single-process identity (the world_size=1 path every
+    single-GPU run takes)."""
+    from es_pytorch_amd.parallel.comm import Comm
+    comm = Comm(torch.device("cpu"))
+    assert comm.size == 1 and comm.rank == 0
+
+    rows = torch.arange(6, dtype=torch.float64).reshape(2, 3)
+    assert torch.equal(comm.allgather_rows(rows), rows)
+
+    t = torch.ones(3)
+    comm.allreduce_sum_(t)
+    assert torch.equal(t, torch.ones(3))
+
+    assert comm.allreduce_scalar(2.5) == 2.5
+    assert comm.broadcast_obj({"a": 1}) == {"a": 1}
+    b = torch.tensor([4.0, 5.0])
+    comm.broadcast_tensor_(b)
+    assert torch.equal(b, torch.tensor([4.0, 5.0]))
+    assert comm.allgather_obj("x") == ["x"]
+    comm.barrier()  # no-op, must not hang
