@@ -36,6 +36,8 @@ def test_bench_cpu_single():
         assert k in out, k
     assert out["value"] > 0 and out["scaling"] == "weak"
     assert out["config"]["global_batch"] == 4
+    # encoding transparency: the perturbation format is always reported
+    assert out["config"]["eps_encoding"] in ("bf16", "e4m3")
 
 
 @pytest.mark.timeout(600)
